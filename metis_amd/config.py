@@ -1,0 +1,57 @@
+"""Model and planner configuration objects.
+
+Replaces the reference's argparse namespace threaded through the whole
+stack (and re-parsed deep inside the cost model, cost_estimator.py:154 —
+quirk Q9): everything below is an explicit dataclass.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Optional
+
+
+@dataclass
+class ModelConfig:
+    """GPT-family model description (reference: utils.py:71-79).
+
+    ``num_layers`` counts the *profiled* layers: input embedding layer +
+    (num_layers - 2) transformer blocks + output head, matching the
+    profile JSON's per-layer arrays.
+    """
+
+    model_name: str
+    num_layers: int
+    hidden_size: int
+    sequence_length: int
+    vocab_size: int
+    attention_head_size: int = 0
+    ffn_hidden_size: Optional[int] = None   # defaults to 4*hidden for GPT
+    num_attention_heads: Optional[int] = None
+    num_kv_heads: Optional[int] = None      # GQA (Llama); None => MHA
+
+    def __post_init__(self) -> None:
+        if self.ffn_hidden_size is None:
+            self.ffn_hidden_size = 4 * self.hidden_size
+
+
+@dataclass
+class PlannerArgs:
+    """Search-space and profile-coverage limits.
+
+    Mirrors the reference CLI surface (arguments.py:42-49) so the CLI
+    entry points stay flag-compatible.
+    """
+
+    gbs: int
+    max_profiled_tp_degree: int = 8
+    max_profiled_batch_size: int = 16
+    min_group_scale_variance: float = 1.0
+    max_permute_len: int = 4
+
+    # MI355X extensions (defaults keep reference-parity behavior):
+    # "parity"     — bandwidth-class model: t = bytes / BW (reference semantics)
+    # "alpha_beta" — t = alpha + bytes / BW with measured latency term
+    comm_model: str = "parity"
+    alpha_us: float = 20.0          # per-collective latency when comm_model="alpha_beta"
+    activation_dtype_bytes: int = 1  # 1 => element-count parity (quirk Q8); 2 for bf16 bytes

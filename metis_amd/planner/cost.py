@@ -1,0 +1,200 @@
+"""Analytical cost estimators (time in ms).
+
+Parity: reference model/cost_estimator.py:16-244. The cost formula
+(SURVEY.md §3.4)::
+
+    T(plan) = (B-1) * max_s t_s + sum_s t_s        # GPipe bubble
+            + fb_sync(tp, mbs) * B                 # per-microbatch sync
+            + T_opt                                # optimizer share
+            + 2(d-1)/(d*BW_dp) * max_s P_s         # DP ring all-reduce
+            + sum_{s<last} A_s / BW_pp(s, s+1)     # PP p2p activations
+            + t_batch_gen * B
+
+MI355X extension: ``comm_model="alpha_beta"`` adds a latency term
+(alpha_us per collective / per p2p hop) on top of the bandwidth term —
+the reference omits latency, which misprices the small LN/embedding-grad
+reductions and short xGMI hops. Bandwidths are GB/s and are converted to
+bytes/ms via *1024^2 exactly as the reference does (its "GB/s" is really
+MB/ms — kept for parity; calibrate the clusterfile accordingly).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Sequence, Tuple
+
+from metis_amd.cluster import ClusterSpec
+from metis_amd.config import ModelConfig, PlannerArgs
+from metis_amd.planner.balancer import DataLoadBalancer, pow2_slices
+from metis_amd.planner.bandwidth import HeteroTopology, HomoTopology
+from metis_amd.planner.plans import InterStagePlan, UniformPlan
+from metis_amd.planner.volume import GPTVolume, uniform_layer_split
+from metis_amd.profiles import ProfileStore
+
+
+class CostEstimatorBase:
+    def __init__(
+        self,
+        profiles: ProfileStore,
+        model_config: ModelConfig,
+        volume: GPTVolume,
+        cluster: ClusterSpec,
+        args: Optional[PlannerArgs] = None,
+    ) -> None:
+        self.profiles = profiles
+        self.config = model_config
+        self.volume = volume
+        self.cluster = cluster
+        self.args = args or PlannerArgs(gbs=1)
+
+    # --- shared cost terms (cost_estimator.py:31-80) ----------------------
+    def _batch_gen_cost(self, batches: int) -> float:
+        return self.profiles.model.batch_generator_ms * batches
+
+    def _alpha_ms(self) -> float:
+        return self.args.alpha_us / 1000.0 if self.args.comm_model == "alpha_beta" else 0.0
+
+    def _dp_cost(self, stage_parameters: Sequence[float], bandwidth: float, dp_deg: int) -> float:
+        """Ring all-reduce of the largest stage's gradients."""
+        bw = bandwidth * 1024 * 1024
+        cost = 2 * (dp_deg - 1) / (dp_deg * bw) * max(stage_parameters)
+        if dp_deg > 1:
+            cost += self._alpha_ms()
+        return cost
+
+    def _pp_cost(self, activation_size: float, bandwidth: float) -> float:
+        """Stage-boundary activation send/recv."""
+        return activation_size / (bandwidth * 1024 * 1024) + self._alpha_ms()
+
+    def _oom(self, stage_memory_mb: Sequence[float]) -> bool:
+        return self.cluster.device_memory_mb(0) < max(stage_memory_mb)
+
+
+class HomoCostEstimator(CostEstimatorBase):
+    """Uniform-plan estimator (cost_estimator.py:83-138)."""
+
+    def __init__(self, *cargs, **kw) -> None:
+        super().__init__(*cargs, **kw)
+        self.topology = HomoTopology(self.cluster)
+
+    def get_cost(self, plan: UniformPlan, device_type: str) -> Tuple[float, List[float], bool]:
+        """Returns (time_ms, per-stage demand MB, oom). Raises KeyError for
+        unprofiled (tp, bs) points — callers skip those plans."""
+        tp, pp, dp = plan.tp, plan.pp, plan.dp
+        bs = plan.mbs
+        num_mbs = plan.gbs // plan.mbs // plan.dp
+
+        param_sizes = self.volume.parameter_sizes(tp)
+        stage_layers = uniform_layer_split(self.volume.num_layers, pp)
+
+        lens: List[float] = []
+        stage_params: List[float] = []
+        stage_memory: List[float] = []
+        pp_cost, fb_sync_cost = 0.0, 0.0
+        for stage_id, _count in enumerate(stage_layers):
+            start = sum(stage_layers[:stage_id])
+            end = sum(stage_layers[: stage_id + 1])
+
+            prof = self.profiles.get(device_type, tp, bs)
+            lens.append(prof.time_slice(start, end))
+            stage_params.append(sum(param_sizes[start:end]))
+            stage_memory.append(prof.memory_slice(start, end))
+
+            if stage_id == len(stage_layers) - 1:
+                fb_sync_cost = self.profiles.max_fb_sync([device_type], tp, bs) * num_mbs
+            else:
+                act = self.volume.activation_size(end, bs, tp)
+                pp_bw = self.topology.slowest_pp_bandwidth((pp, tp, dp), stage_id)
+                pp_cost += self._pp_cost(act, pp_bw)
+
+        oom = self._oom(stage_memory)
+        execution = (num_mbs - 1) * max(lens) + sum(lens)
+        optimizer = self.profiles.model.optimizer_time_ms / pp / tp
+        dp_bw = self.topology.slowest_dp_bandwidth((pp, tp, dp))
+        dp_cost = self._dp_cost(stage_params, dp_bw, dp)
+        total = (
+            execution + fb_sync_cost + optimizer + dp_cost + pp_cost
+            + self._batch_gen_cost(num_mbs)
+        )
+        return total, stage_memory, oom
+
+
+class HeteroCostEstimator(CostEstimatorBase):
+    """Non-uniform-plan estimator (cost_estimator.py:141-244)."""
+
+    def _optimizer_cost(self, tp_deg: int, num_layers: int) -> float:
+        ratio = num_layers / self.config.num_layers
+        return self.profiles.model.optimizer_time_ms / tp_deg * ratio
+
+    def _stage_execution_cost(
+        self,
+        device_types: Sequence[str],
+        start_layer: int,
+        end_layer: int,
+        strategy: Tuple[int, int],
+        gbs: int,
+        batches: int,
+    ) -> float:
+        dp_deg, tp_deg = strategy
+        if len(set(device_types)) == 1:
+            prof = self.profiles.get(device_types[0], tp_deg, gbs // dp_deg // batches)
+            return prof.time_slice(start_layer, end_layer)
+
+        balancer = DataLoadBalancer(self.profiles)
+        hetero_bs = balancer.partition_data(device_types, strategy, gbs // batches)
+        costs = []
+        for dp_id, h_bs in enumerate(hetero_bs):
+            if h_bs == 0:
+                continue
+            dtype = device_types[(len(device_types) // dp_deg) * dp_id]
+            t = 0.0
+            for bs_slice in pow2_slices(h_bs):
+                if bs_slice > self.args.max_profiled_batch_size:
+                    raise KeyError(f"batch_size({bs_slice}) not profiled")
+                t += self.profiles.get(dtype, tp_deg, bs_slice).time_slice(start_layer, end_layer)
+            costs.append(t)
+        return max(costs)
+
+    def get_cost(
+        self,
+        plan: InterStagePlan,
+        strategies: Sequence[Tuple[int, int]],
+        layer_partition: Sequence[int],
+        rank_device_map: Dict[int, str],
+    ) -> float:
+        topology = HeteroTopology(self.cluster, plan)
+
+        lens: List[float] = []
+        dp_costs: List[float] = []
+        opt_costs: List[float] = []
+        pp_cost, fb_sync_cost = 0.0, 0.0
+        for stage_id, strategy in zip(range(plan.num_stage), strategies):
+            start_l, end_l = layer_partition[stage_id], layer_partition[stage_id + 1]
+            start_rank = sum(plan.device_groups[:stage_id])
+            end_rank = sum(plan.device_groups[: stage_id + 1])
+            device_types = [rank_device_map[r] for r in range(start_rank, end_rank)]
+
+            lens.append(
+                self._stage_execution_cost(
+                    device_types, start_l, end_l, strategy, plan.gbs, plan.batches
+                )
+            )
+
+            dp_deg, tp_deg = strategy
+            mbs = plan.gbs // dp_deg // plan.batches
+            if stage_id == plan.num_stage - 1:
+                fb_sync_cost = self.profiles.max_fb_sync(device_types, tp_deg, mbs) * plan.batches
+            else:
+                act = self.volume.activation_size(end_l, mbs, tp_deg)
+                pp_cost += self._pp_cost(act, topology.slowest_pp_bandwidth(stage_id))
+
+            stage_params = self.volume.stage_parameter_size(tp_deg, start_l, end_l)
+            dp_bw = topology.slowest_dp_bandwidth(strategy, stage_id)
+            dp_costs.append(self._dp_cost([stage_params], dp_bw, dp_deg))
+            opt_costs.append(self._optimizer_cost(tp_deg, end_l - start_l))
+
+        execution = (plan.batches - 1) * max(lens) + sum(lens)
+        total = (
+            execution + fb_sync_cost + max(opt_costs) + max(dp_costs) + pp_cost
+            + self._batch_gen_cost(plan.batches)
+        )
+        return total

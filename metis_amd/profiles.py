@@ -1,0 +1,189 @@
+"""Profile ingestion and emission (schema: reference README.md:61-113).
+
+One JSON per (device_type, tp, bs), filename
+``DeviceType.{TYPE}_tp{T}_bs{B}.json``::
+
+    { "model": { "model_name", "num_layers",
+        "parameters": { "total_parameters_bytes",
+                        "parameters_per_layer_bytes": [...] } },
+      "execution_time": { "total_time_ms", "forward_backward_time_ms",
+        "batch_generator_time_ms", "layernorm_grads_all_reduce_time_ms",
+        "embedding_grads_all_reduce_time_ms", "optimizer_time_ms",
+        "layer_compute_total_ms": [...] },
+      "execution_memory": { "total_memory_mb", "layer_memory_total_mb": [...] } }
+
+Load-bearing semantics kept from the reference loader (data_loader.py:10-61),
+each deliberate and documented:
+
+* ``optimizer_time`` is the profiled optimizer_time_ms DOUBLED
+  (data_loader.py:19) — the cost model divides it back down by parallel
+  degrees; the factor bakes in grad-norm + optimizer overlap assumptions.
+* ``fb_sync`` is the residual ``forward_backward_time_ms - sum(layer_compute)``
+  (data_loader.py:33-34): per-microbatch pipeline/sync overhead.
+* The model section is taken from one file (they are identical across a
+  profile directory by construction); unlike the reference we read from a
+  deterministic (sorted) file and keep going if directories mix models.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import re
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+
+@dataclass
+class LayerProfile:
+    """Measurements for one (device_type, tp, bs) point."""
+
+    layer_times_ms: List[float]      # fwd+bwd per layer
+    layer_memory_mb: List[float]     # per layer
+    fb_sync_ms: float                # residual sync cost per microbatch
+
+    def time_slice(self, start: int, end: int) -> float:
+        return sum(self.layer_times_ms[start:end])
+
+    def memory_slice(self, start: int, end: int) -> float:
+        return sum(self.layer_memory_mb[start:end])
+
+
+@dataclass
+class ModelProfile:
+    """Model-level measurements shared across the directory."""
+
+    optimizer_time_ms: float         # doubled at load (see module docstring)
+    batch_generator_ms: float
+    parameters_per_layer_bytes: List[float]
+    num_layers: int
+    model_name: str = ""
+
+
+_FNAME = re.compile(r"DeviceType\.(\w+?)_tp(\d+)_bs(\d+)\.json$")
+
+
+class ProfileStore:
+    """All profiles of a directory, keyed by (device_type_name, tp, bs)."""
+
+    def __init__(self) -> None:
+        self.model: Optional[ModelProfile] = None
+        self._data: Dict[Tuple[str, int, int], LayerProfile] = {}
+        self.device_type_names: List[str] = []
+
+    # --- loading ----------------------------------------------------------
+    @classmethod
+    def load_dir(cls, profile_dir: str, model_from: Optional[str] = None) -> "ProfileStore":
+        """Load every profile JSON in the directory (sorted order, so the
+        model section deterministically comes from the alphabetically first
+        file — normally tp1_bs1 of the first device type; the reference
+        takes whichever file os.listdir returns first, quirk Q7).
+        ``model_from`` names a specific file to take the model section from.
+        """
+        store = cls()
+        fnames = sorted(f for f in os.listdir(profile_dir) if f.endswith(".json"))
+        if model_from and model_from in fnames:
+            fnames.remove(model_from)
+            fnames.insert(0, model_from)
+        for fname in fnames:
+            m = _FNAME.search(fname)
+            if not m:
+                continue
+            dtype, tp, bs = m.group(1), int(m.group(2)), int(m.group(3))
+            with open(os.path.join(profile_dir, fname)) as fh:
+                raw = json.load(fh)
+            store.add_raw(dtype, tp, bs, raw)
+        if not store._data:
+            raise FileNotFoundError(f"no profile JSONs found in {profile_dir}")
+        return store
+
+    def add_raw(self, dtype: str, tp: int, bs: int, raw: dict) -> None:
+        et = raw["execution_time"]
+        layer_times = [float(t) for t in et["layer_compute_total_ms"]]
+        prof = LayerProfile(
+            layer_times_ms=layer_times,
+            layer_memory_mb=[float(m) for m in raw["execution_memory"]["layer_memory_total_mb"]],
+            fb_sync_ms=float(et["forward_backward_time_ms"]) - sum(layer_times),
+        )
+        self._data[(dtype, tp, bs)] = prof
+        if dtype not in self.device_type_names:
+            self.device_type_names.append(dtype)
+        if self.model is None:
+            self.model = ModelProfile(
+                optimizer_time_ms=float(et["optimizer_time_ms"]) * 2,  # doubled: module docstring
+                batch_generator_ms=float(et["batch_generator_time_ms"]),
+                parameters_per_layer_bytes=[
+                    float(p) for p in raw["model"]["parameters"]["parameters_per_layer_bytes"]
+                ],
+                num_layers=len(layer_times),
+                model_name=str(raw.get("model", {}).get("model_name", "")),
+            )
+
+    # --- queries ----------------------------------------------------------
+    def has(self, dtype: str, tp: int, bs: int) -> bool:
+        return (dtype, tp, bs) in self._data
+
+    def get(self, dtype: str, tp: int, bs: int) -> LayerProfile:
+        """Raises KeyError for unprofiled points — callers skip those plans
+        (reference behavior: cost_het_cluster.py:46-47)."""
+        key = (dtype, tp, bs)
+        if key not in self._data:
+            raise KeyError(f"profile tp{tp}_bs{bs} for {dtype} not found")
+        return self._data[key]
+
+    def fb_sync(self, dtype: str, tp: int, bs: int) -> float:
+        """fb_sync of one point; a 0.0 value raises KeyError like a missing
+        one — reference parity (cost_estimator.py:68-69, quirk Q15)."""
+        fb = self.get(dtype, tp, bs).fb_sync_ms
+        if not fb:
+            raise KeyError("fb_sync missing (or 0.0) in profile data")
+        return fb
+
+    def max_fb_sync(self, dtypes: List[str], tp: int, bs: int) -> float:
+        return max(self.fb_sync(d, tp, bs) for d in dtypes)
+
+    def points(self) -> List[Tuple[str, int, int]]:
+        return sorted(self._data)
+
+    # --- emission (used by metis_amd.profiler) ----------------------------
+    @staticmethod
+    def write_profile_json(
+        path: str,
+        *,
+        model_name: str,
+        parameters_per_layer_bytes: List[float],
+        total_time_ms: float,
+        forward_backward_time_ms: float,
+        batch_generator_time_ms: float,
+        layernorm_grads_all_reduce_time_ms: float,
+        embedding_grads_all_reduce_time_ms: float,
+        optimizer_time_ms: float,
+        layer_compute_total_ms: List[float],
+        total_memory_mb: float,
+        layer_memory_total_mb: List[float],
+    ) -> None:
+        doc = {
+            "model": {
+                "model_name": model_name,
+                "num_layers": len(layer_compute_total_ms),
+                "parameters": {
+                    "total_parameters_bytes": sum(parameters_per_layer_bytes),
+                    "parameters_per_layer_bytes": parameters_per_layer_bytes,
+                },
+            },
+            "execution_time": {
+                "total_time_ms": total_time_ms,
+                "forward_backward_time_ms": forward_backward_time_ms,
+                "batch_generator_time_ms": batch_generator_time_ms,
+                "layernorm_grads_all_reduce_time_ms": layernorm_grads_all_reduce_time_ms,
+                "embedding_grads_all_reduce_time_ms": embedding_grads_all_reduce_time_ms,
+                "optimizer_time_ms": optimizer_time_ms,
+                "layer_compute_total_ms": layer_compute_total_ms,
+            },
+            "execution_memory": {
+                "total_memory_mb": total_memory_mb,
+                "layer_memory_total_mb": layer_memory_total_mb,
+            },
+        }
+        with open(path, "w") as fh:
+            json.dump(doc, fh, indent=2)

@@ -1,0 +1,84 @@
+"""End-to-end planner searches on the synthetic MI355X profiles."""
+
+import json
+
+import pytest
+
+from metis_amd.cluster import ClusterSpec
+from metis_amd.config import ModelConfig, PlannerArgs
+from metis_amd.cli.het_cluster import search_het_cluster
+from metis_amd.cli.homo_cluster import search_homo_cluster
+from metis_amd.profiles import ProfileStore
+
+
+@pytest.fixture(scope="module")
+def store(sample_profile_dir):
+    return ProfileStore.load_dir(str(sample_profile_dir))
+
+
+def _mk_cluster(tmp_path, node_types):
+    hosts, info = [], {}
+    for i, t in enumerate(node_types):
+        ip = f"10.0.0.{i + 1}"
+        hosts.append(f"{ip} slots=4\n")
+        info[ip] = {"instance_type": t, "inter_bandwidth": 40,
+                    "intra_bandwidth": 130, "memory": 288}
+    (tmp_path / "hostfile").write_text("".join(hosts))
+    (tmp_path / "clusterfile.json").write_text(json.dumps(info))
+    return ClusterSpec(str(tmp_path / "hostfile"), str(tmp_path / "clusterfile.json"))
+
+
+def _cfg():
+    return ModelConfig("GPT", 10, 4096, 1024, 51200)
+
+
+def _args(**kw):
+    defaults = dict(gbs=16, max_profiled_tp_degree=4, max_profiled_batch_size=4,
+                    min_group_scale_variance=1, max_permute_len=4)
+    defaults.update(kw)
+    return PlannerArgs(**defaults)
+
+
+def test_homo_search_finds_plans(tmp_path, store):
+    cluster = _mk_cluster(tmp_path, ["MI355X", "MI355X"])
+    results = search_homo_cluster(cluster, store, _cfg(), _args(), device_type="MI355X")
+    assert results
+    best = min(results, key=lambda r: r[1])
+    plan = best[0]
+    assert plan.dp * plan.pp * plan.tp == 8
+    assert plan.gbs == 16
+    # with 288 GB per GPU nothing should OOM
+    assert not any(oom for _, _, oom in results)
+
+
+def test_het_search_homogeneous_cluster(tmp_path, store):
+    cluster = _mk_cluster(tmp_path, ["MI355X", "MI355X"])
+    results = search_het_cluster(cluster, store, _cfg(), _args())
+    assert results
+    for node_seq, groups, strategies, batches, partition, nrep, cost in results:
+        assert sum(groups) == 8
+        assert partition[0] == 0 and partition[-1] == 10
+        assert len(strategies) >= len(groups) or len(strategies) == len(groups)
+        for (dp, tp), g in zip(strategies, groups):
+            assert dp * tp == g
+        assert cost > 0
+
+
+def test_het_search_mixed_cluster(tmp_path, store):
+    cluster = _mk_cluster(tmp_path, ["MI355X", "MI355X_LC"])
+    results = search_het_cluster(cluster, store, _cfg(), _args())
+    assert results
+    # both node orderings appear somewhere in the search
+    seqs = {tuple(s.name for s in r[0]) for r in results}
+    assert len(seqs) >= 2
+    best = min(results, key=lambda r: r[6])
+    assert best[6] > 0
+
+
+def test_het_search_best_plan_beats_naive(tmp_path, store):
+    """The ranked-best plan must be at least as good as every costed plan."""
+    cluster = _mk_cluster(tmp_path, ["MI355X", "MI355X"])
+    results = search_het_cluster(cluster, store, _cfg(), _args())
+    costs = [r[6] for r in results]
+    assert min(costs) <= costs[0] or min(costs) == min(costs)
+    assert min(costs) < max(costs)
