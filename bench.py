@@ -1,0 +1,99 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: best-plan training iteration time, GPT-3 2.7B.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W` runs the
+flagship training step on N GPUs of one node (launched via
+torch.distributed.run for N > 1, one rank per GPU over RCCL). Rank 0
+prints ONE JSON line with the measured iteration time.
+
+Metric (BASELINE.json): best-plan measured iteration time (ms) for
+GPT-3 2.7B on synthetic data / random-init weights, bf16. Weak scaling:
+per-GPU batch is fixed, so gbs grows with N.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+
+import torch
+import torch.distributed as dist
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from metis_amd.models.gpt import MODEL_SPECS  # noqa: E402
+from metis_amd.runtime.comm import init_parallel  # noqa: E402
+from metis_amd.runtime.runner import PlanRunner  # noqa: E402
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", default="gpt3-2.7b", choices=sorted(MODEL_SPECS))
+    p.add_argument("--per-gpu-batch", type=int, default=4,
+                   help="sequences per GPU per step (weak scaling)")
+    p.add_argument("--mbs", type=int, default=4, help="microbatch size")
+    p.add_argument("--tp", type=int, default=1)
+    p.add_argument("--pp", type=int, default=1)
+    return p.parse_args()
+
+
+def main() -> None:
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    n_gpus = max(world, args.gpus if world == 1 else world)
+    spec = MODEL_SPECS[args.model]
+
+    dp = n_gpus // (args.tp * args.pp)
+    assert dp * args.tp * args.pp == n_gpus, "tp*pp must divide --gpus"
+    gbs = args.per_gpu_batch * n_gpus // (args.tp * args.pp)
+    mbs = min(args.mbs, max(gbs // dp, 1))
+
+    ctx = init_parallel(dp=dp, tp=args.tp, pp=args.pp)
+    runner = PlanRunner(spec, ctx, mbs=mbs, gbs=gbs)
+
+    ms = runner.timed_steps(args.steps, args.warmup)
+
+    # per-rank max: the slowest rank defines the iteration time
+    if dist.is_initialized():
+        t = torch.tensor([ms], dtype=torch.float64,
+                         device=ctx.device if ctx.device else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        ms = float(t.item())
+
+    if ctx.rank == 0:
+        tokens_per_step = gbs * spec.seq_length
+        result = {
+            "metric": "best_plan_iter_time_ms",
+            "value": ms,
+            "unit": "ms",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms,
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": gbs,
+                "seq_len": spec.seq_length,
+                "parallelism": f"dp{dp}_tp{args.tp}_pp{args.pp}",
+                "microbatch": mbs,
+                "tokens_per_s": tokens_per_step / (ms / 1000.0),
+            },
+        }
+        print(json.dumps(result))
+
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,296 @@
+"""GPT model family with tensor-parallel sharding over RCCL.
+
+Megatron-style TP: QKV / fc1 are column-parallel (f: identity fwd,
+all-reduce bwd), proj / fc2 are row-parallel (g: all-reduce fwd, identity
+bwd); one all-reduce per attention block and one per MLP block in each
+direction. LayerNorm runs the fused gfx950 kernel (metis_amd.ops); GEMMs
+go through torch.matmul (hipBLASLt on ROCm); attention uses
+scaled_dot_product_attention until the hand-written flash kernel wires in.
+
+The layer numbering matches the Metis profile convention: layer 0 =
+embedding, 1..n-2 = transformer blocks, n-1 = LM head (+ final norm).
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+import torch.nn.functional as F
+
+from metis_amd.ops import LayerNorm
+
+
+@dataclass(frozen=True)
+class GPTModelSpec:
+    name: str
+    hidden_size: int
+    num_layers: int          # transformer blocks
+    num_heads: int
+    vocab_size: int
+    seq_length: int
+    ffn_hidden_size: Optional[int] = None
+    num_kv_heads: Optional[int] = None
+
+    @property
+    def ffn(self) -> int:
+        return self.ffn_hidden_size or 4 * self.hidden_size
+
+    @property
+    def head_dim(self) -> int:
+        return self.hidden_size // self.num_heads
+
+    @property
+    def profile_num_layers(self) -> int:
+        """Layer count in the Metis profile convention (embed + blocks + head)."""
+        return self.num_layers + 2
+
+    def num_parameters(self) -> int:
+        h, v = self.hidden_size, self.vocab_size
+        per_block = 4 * h * h + 2 * h * self.ffn + 4 * h + self.ffn + h + 2 * h
+        return v * h + self.seq_length * h + self.num_layers * per_block + 2 * h
+
+
+MODEL_SPECS = {
+    "gpt2-small": GPTModelSpec("gpt2-small", 768, 12, 12, 51200, 1024),
+    "gpt3-1.3b": GPTModelSpec("gpt3-1.3b", 2048, 24, 16, 51200, 2048),
+    "gpt3-2.7b": GPTModelSpec("gpt3-2.7b", 2560, 32, 32, 51200, 2048),
+    "gpt3-6.7b": GPTModelSpec("gpt3-6.7b", 4096, 32, 32, 51200, 2048),
+    # 10-layer GPT-3-shaped config matching the reference's bundled profiles
+    "gpt3-10l": GPTModelSpec("gpt3-10l", 4096, 8, 32, 51200, 1024),
+}
+
+
+# --- tensor-parallel autograd collectives ---------------------------------
+class _CopyToTP(torch.autograd.Function):
+    """f: identity forward, grad all-reduce backward (column-parallel in)."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        return x
+
+    @staticmethod
+    def backward(ctx, grad):
+        if ctx.group is not None and dist.get_world_size(ctx.group) > 1:
+            grad = grad.contiguous()
+            dist.all_reduce(grad, group=ctx.group)
+        return grad, None
+
+
+class _ReduceFromTP(torch.autograd.Function):
+    """g: all-reduce forward, identity backward (row-parallel out)."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        if group is not None and dist.get_world_size(group) > 1:
+            x = x.contiguous()
+            dist.all_reduce(x, group=group)
+        return x
+
+    @staticmethod
+    def backward(ctx, grad):
+        return grad, None
+
+
+class _VocabParallelCrossEntropy(torch.autograd.Function):
+    """Cross entropy over vocab-sharded logits without gathering them
+    (Megatron-style): three small all-reduces (max, sum-exp, target logit)
+    instead of moving the [tokens, vocab] tensor."""
+
+    @staticmethod
+    def forward(ctx, logits_shard, labels, group):
+        rank = dist.get_rank(group)
+        vp = logits_shard.size(-1)
+        vocab_start = rank * vp
+
+        m = logits_shard.max(dim=-1).values
+        dist.all_reduce(m, op=dist.ReduceOp.MAX, group=group)
+        exp = torch.exp(logits_shard - m[:, None])
+        sumexp = exp.sum(dim=-1)
+        dist.all_reduce(sumexp, group=group)
+
+        local = (labels >= vocab_start) & (labels < vocab_start + vp)
+        idx = (labels - vocab_start).clamp(0, vp - 1)
+        target_logit = torch.where(
+            local, logits_shard.gather(1, idx[:, None]).squeeze(1),
+            torch.zeros_like(m),
+        )
+        dist.all_reduce(target_logit, group=group)
+
+        loss = (torch.log(sumexp) + m - target_logit).mean()
+        ctx.save_for_backward(exp, sumexp, local, idx)
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        exp, sumexp, local, idx = ctx.saved_tensors
+        n = exp.size(0)
+        softmax = exp / sumexp[:, None]
+        softmax.scatter_add_(
+            1, idx[:, None],
+            torch.where(local, -torch.ones_like(sumexp), torch.zeros_like(sumexp))[:, None],
+        )
+        return softmax * (grad_out / n), None, None
+
+
+def _init_linear(weight: torch.Tensor, fan_in: int) -> None:
+    std = 1.0 / math.sqrt(fan_in)
+    nn.init.normal_(weight, mean=0.0, std=std)
+
+
+class ColumnParallelLinear(nn.Module):
+    """y_shard = x @ W_shard^T + b_shard, W sharded over output dim."""
+
+    def __init__(self, in_features: int, out_features: int, tp: int, dtype):
+        super().__init__()
+        assert out_features % tp == 0
+        self.out_per_rank = out_features // tp
+        self.weight = nn.Parameter(torch.empty(self.out_per_rank, in_features, dtype=dtype))
+        self.bias = nn.Parameter(torch.zeros(self.out_per_rank, dtype=dtype))
+        _init_linear(self.weight, in_features)
+
+    def forward(self, x: torch.Tensor, tp_group) -> torch.Tensor:
+        x = _CopyToTP.apply(x, tp_group)
+        return F.linear(x, self.weight, self.bias)
+
+
+class RowParallelLinear(nn.Module):
+    """y = all_reduce(x_shard @ W_shard^T) + b, W sharded over input dim."""
+
+    def __init__(self, in_features: int, out_features: int, tp: int, dtype):
+        super().__init__()
+        assert in_features % tp == 0
+        self.in_per_rank = in_features // tp
+        self.weight = nn.Parameter(torch.empty(out_features, self.in_per_rank, dtype=dtype))
+        self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype))
+        _init_linear(self.weight, in_features)
+
+    def forward(self, x: torch.Tensor, tp_group) -> torch.Tensor:
+        y = F.linear(x, self.weight)
+        y = _ReduceFromTP.apply(y, tp_group)
+        return y + self.bias
+
+
+class GPTBlock(nn.Module):
+    def __init__(self, spec: GPTModelSpec, tp: int, dtype):
+        super().__init__()
+        h = spec.hidden_size
+        assert spec.num_heads % tp == 0, "num_heads must divide by tp"
+        self.heads_per_rank = spec.num_heads // tp
+        self.head_dim = spec.head_dim
+
+        self.ln_attn = LayerNorm(h)
+        self.qkv = ColumnParallelLinear(h, 3 * h, tp, dtype)
+        self.proj = RowParallelLinear(h, h, tp, dtype)
+        self.ln_mlp = LayerNorm(h)
+        self.fc1 = ColumnParallelLinear(h, spec.ffn, tp, dtype)
+        self.fc2 = RowParallelLinear(spec.ffn, h, tp, dtype)
+
+    def forward(self, x: torch.Tensor, tp_group) -> torch.Tensor:
+        b, s, _ = x.shape
+        residual = x
+        y = self.ln_attn(x)
+        qkv = self.qkv(y, tp_group)
+        qkv = qkv.view(b, s, self.heads_per_rank, 3 * self.head_dim)
+        q, k, v = qkv.chunk(3, dim=-1)
+        q = q.transpose(1, 2)  # [b, heads, s, d]
+        k = k.transpose(1, 2)
+        v = v.transpose(1, 2)
+        attn = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        attn = attn.transpose(1, 2).reshape(b, s, -1)
+        x = residual + self.proj(attn, tp_group)
+
+        residual = x
+        y = self.ln_mlp(x)
+        y = self.fc1(y, tp_group)
+        y = F.gelu(y, approximate="tanh")
+        x = residual + self.fc2(y, tp_group)
+        return x
+
+
+class GPTModel(nn.Module):
+    """A pipeline-stage slice of the GPT model.
+
+    ``layer_range`` selects profile-convention layers [start, end) out of
+    [0, num_layers+2): owning layer 0 adds the embeddings, owning the last
+    layer adds the final norm + LM head.
+    """
+
+    def __init__(
+        self,
+        spec: GPTModelSpec,
+        tp: int = 1,
+        dtype: torch.dtype = torch.bfloat16,
+        layer_range: Optional[tuple] = None,
+        tp_group=None,
+    ):
+        super().__init__()
+        self.spec = spec
+        self.tp = tp
+        self.tp_group = tp_group
+        total = spec.profile_num_layers
+        start, end = layer_range if layer_range is not None else (0, total)
+        assert 0 <= start < end <= total
+        self.has_embedding = start == 0
+        self.has_head = end == total
+
+        h = spec.hidden_size
+        if self.has_embedding:
+            self.wte = nn.Embedding(spec.vocab_size, h, dtype=dtype)
+            self.wpe = nn.Embedding(spec.seq_length, h, dtype=dtype)
+            nn.init.normal_(self.wte.weight, std=0.02)
+            nn.init.normal_(self.wpe.weight, std=0.02)
+
+        block_start = max(start - 1, 0)
+        block_end = min(end, total - 1) - 1
+        self.blocks = nn.ModuleList(
+            GPTBlock(spec, tp, dtype) for _ in range(max(block_end - block_start, 0))
+        )
+
+        if self.has_head:
+            self.ln_final = LayerNorm(h)
+            self.head = ColumnParallelLinear(h, spec.vocab_size, tp, dtype)
+
+    def forward(
+        self, x: torch.Tensor, labels: Optional[torch.Tensor] = None
+    ) -> torch.Tensor:
+        """x: token ids [b, s] on the first stage, hidden states elsewhere.
+        Returns the loss when this stage has the head and labels are given,
+        otherwise the stage's output hidden states."""
+        if self.has_embedding:
+            b, s = x.shape
+            pos = torch.arange(s, device=x.device)
+            x = self.wte(x) + self.wpe(pos)[None, :, :]
+
+        for block in self.blocks:
+            x = block(x, self.tp_group)
+
+        if self.has_head:
+            x = self.ln_final(x)
+            logits = self.head(x, self.tp_group)  # [b, s, vocab/tp]
+            if labels is not None:
+                return self._loss(logits, labels)
+            return logits
+        return x
+
+    def _loss(self, logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+        flat = logits.float().view(-1, logits.size(-1))
+        labels = labels.reshape(-1)
+        if self.tp_group is not None and dist.get_world_size(self.tp_group) > 1:
+            return _VocabParallelCrossEntropy.apply(flat, labels, self.tp_group)
+        return F.cross_entropy(flat, labels)
+
+    def layer_parameter_bytes(self) -> List[float]:
+        """parameters_per_layer_bytes for the profile JSON (full model)."""
+        spec = self.spec
+        el = 2  # bf16
+        h, v_ = spec.hidden_size, spec.vocab_size
+        embed = (v_ * h + spec.seq_length * h) * el
+        per_block = (4 * h * h + 2 * h * spec.ffn + 4 * h + spec.ffn + h + 2 * h) * el
+        head = (v_ * h + 2 * h) * el
+        return [float(embed)] + [float(per_block)] * spec.num_layers + [float(head)]
