@@ -13,6 +13,12 @@ void adamw_step(
     torch::Tensor m, torch::Tensor v,
     double lr, double beta1, double beta2, double eps,
     double weight_decay, long step, double grad_scale);
+std::vector<torch::Tensor> attn_fwd(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v, double scale);
+torch::Tensor gemm_bf16(
+    torch::Tensor a, torch::Tensor b_nk, c10::optional<torch::Tensor> bias,
+    long epilogue);
+torch::Tensor mfma_tile_probe(torch::Tensor a, torch::Tensor b);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("layernorm_fwd", &layernorm_fwd,
@@ -20,4 +26,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("layernorm_bwd", &layernorm_bwd,
           "fused LayerNorm backward (dx + dgamma/dbeta)");
     m.def("adamw_step", &adamw_step, "fused AdamW step");
+    m.def("attn_fwd", &attn_fwd,
+          "flash attention forward (causal, GQA) -> (O, LSE)");
+    m.def("gemm_bf16", &gemm_bf16,
+          "MFMA bf16 GEMM A[M,K] @ B[N,K]^T with fused epilogue "
+          "(0=none, 1=bias, 2=bias+gelu)");
+    m.def("mfma_tile_probe", &mfma_tile_probe,
+          "single-wave 16x16x32 MFMA layout probe");
 }

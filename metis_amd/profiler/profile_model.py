@@ -1,0 +1,274 @@
+"""Per-layer GPT profiler -> ``DeviceType.{TYPE}_tp{T}_bs{B}.json``.
+
+The component the reference README prescribes (README.md:142-186) but does
+not ship: per-layer forward/backward times via hipEvent brackets installed
+with module hooks, per-layer memory (parameter + optimizer state bytes +
+measured activation deltas), Megatron-style interval timers for
+forward_backward / batch_generator / grads-all-reduce / optimizer, one
+iteration == one epoch (mbs = gbs = bs, README.md:84).
+
+Single GPU profiles tp=1; ``torchrun --nproc-per-node T`` profiles tp=T
+(rank 0 writes the JSON).
+"""
+
+from __future__ import annotations
+
+import argparse
+import os
+import time
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from metis_amd.models.gpt import GPTModel, GPTModelSpec, MODEL_SPECS
+from metis_amd.ops import FusedAdamW
+from metis_amd.profiles import ProfileStore
+
+
+class _LayerTimer:
+    """fwd/bwd hipEvent brackets around one profile layer (a module)."""
+
+    def __init__(self, module: torch.nn.Module):
+        self.fwd_ms = 0.0
+        self.bwd_ms = 0.0
+        self.act_bytes = 0
+        self._ev = lambda: torch.cuda.Event(enable_timing=True)
+        self._fwd_start = self._ev()
+        self._fwd_end = self._ev()
+        self._bwd_start = self._ev()
+        self._bwd_end = self._ev()
+        self._mem0 = 0
+        self._handles = [
+            module.register_forward_pre_hook(self._on_fwd_pre),
+            module.register_forward_hook(self._on_fwd_post),
+            module.register_full_backward_pre_hook(self._on_bwd_pre),
+            module.register_full_backward_hook(self._on_bwd_post),
+        ]
+
+    def _on_fwd_pre(self, module, args):
+        self._mem0 = torch.cuda.memory_allocated()
+        self._fwd_start.record()
+
+    def _on_fwd_post(self, module, args, output):
+        self._fwd_end.record()
+        self.act_bytes = max(self.act_bytes,
+                             torch.cuda.memory_allocated() - self._mem0)
+
+    def _on_bwd_pre(self, module, grad_output):
+        self._bwd_start.record()
+
+    def _on_bwd_post(self, module, grad_input, grad_output):
+        self._bwd_end.record()
+
+    def collect(self) -> None:
+        self.fwd_ms = self._fwd_start.elapsed_time(self._fwd_end)
+        self.bwd_ms = self._bwd_start.elapsed_time(self._bwd_end)
+
+    def remove(self) -> None:
+        for h in self._handles:
+            h.remove()
+
+
+class _EmbeddingLayer(torch.nn.Module):
+    """Wraps wte+wpe as profile layer 0."""
+
+    def __init__(self, model: GPTModel):
+        super().__init__()
+        self.wte = model.wte
+        self.wpe = model.wpe
+
+    def forward(self, tokens):
+        pos = torch.arange(tokens.size(1), device=tokens.device)
+        return self.wte(tokens) + self.wpe(pos)[None, :, :]
+
+
+class _HeadLayer(torch.nn.Module):
+    """Wraps final LN + LM head + loss as the last profile layer."""
+
+    def __init__(self, model: GPTModel):
+        super().__init__()
+        self.ln_final = model.ln_final
+        self.head = model.head
+        self._model = [model]  # avoid registering the full model
+
+    def forward(self, x, labels):
+        m = self._model[0]
+        x = self.ln_final(x)
+        logits = self.head(x, m.tp_group)
+        return m._loss(logits, labels)
+
+
+def profile_model(
+    spec: GPTModelSpec,
+    bs: int,
+    tp: int = 1,
+    device_type: str = "MI355X",
+    out_dir: str = "profiles/mi355x",
+    warmup: int = 3,
+    iters: int = 10,
+    tp_group=None,
+    seq_length: Optional[int] = None,
+) -> Optional[str]:
+    """Profile one (tp, bs) point; returns the JSON path (rank 0)."""
+    assert torch.cuda.is_available(), "profiler needs a GPU"
+    if seq_length:
+        spec = GPTModelSpec(spec.name, spec.hidden_size, spec.num_layers,
+                            spec.num_heads, spec.vocab_size, seq_length,
+                            spec.ffn_hidden_size, spec.num_kv_heads)
+    dev = torch.device("cuda", torch.cuda.current_device())
+    model = GPTModel(spec, tp=tp, dtype=torch.bfloat16, tp_group=tp_group).to(dev)
+    opt = FusedAdamW(model.parameters(), lr=1e-4)
+
+    # profile layers: [embedding, blocks..., head]
+    embed = _EmbeddingLayer(model)
+    head = _HeadLayer(model)
+    layers: List[torch.nn.Module] = [embed, *model.blocks, head]
+    timers = [_LayerTimer(l) for l in layers]
+
+    def batch():
+        tokens = torch.randint(0, spec.vocab_size, (bs, spec.seq_length), device=dev)
+        return tokens, torch.roll(tokens, -1, 1)
+
+    def one_iter(measure: bool):
+        t_iter0 = time.perf_counter()
+        t0 = time.perf_counter()
+        tokens, labels = batch()
+        torch.cuda.synchronize()
+        batch_ms = (time.perf_counter() - t0) * 1000
+
+        opt.zero_grad()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        x = embed(tokens)
+        for blk in model.blocks:
+            x = blk(x, model.tp_group)
+        loss = head(x, labels)
+        loss.backward()
+        torch.cuda.synchronize()
+        fwd_bwd_ms = (time.perf_counter() - t0) * 1000
+
+        # tied/norm grad all-reduce intervals (TP group), as the schema
+        # prescribes (README.md:80-81) — separate timed collectives
+        ln_ms = emb_ms = 0.0
+        if tp > 1 and tp_group is not None:
+            ln_grads = [m.ln_attn.weight.grad for m in model.blocks]
+            ln_grads += [m.ln_mlp.weight.grad for m in model.blocks]
+            flat = torch.cat([g.reshape(-1) for g in ln_grads if g is not None])
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            dist.all_reduce(flat, group=tp_group)
+            torch.cuda.synchronize()
+            ln_ms = (time.perf_counter() - t0) * 1000
+
+            emb_grad = model.wte.weight.grad
+            if emb_grad is not None:
+                torch.cuda.synchronize()
+                t0 = time.perf_counter()
+                dist.all_reduce(emb_grad, group=tp_group)
+                torch.cuda.synchronize()
+                emb_ms = (time.perf_counter() - t0) * 1000
+
+        t0 = time.perf_counter()
+        opt.step(grad_scale=1.0)
+        torch.cuda.synchronize()
+        opt_ms = (time.perf_counter() - t0) * 1000
+        total_ms = (time.perf_counter() - t_iter0) * 1000
+
+        if not measure:
+            return None
+        for t in timers:
+            t.collect()
+        return {
+            "batch": batch_ms, "fwd_bwd": fwd_bwd_ms, "opt": opt_ms,
+            "ln_ar": ln_ms, "emb_ar": emb_ms, "total": total_ms,
+            "layers": [t.fwd_ms + t.bwd_ms for t in timers],
+        }
+
+    for _ in range(warmup):
+        one_iter(measure=False)
+
+    acc: Dict[str, float] = {}
+    layer_acc = [0.0] * len(layers)
+    for _ in range(iters):
+        m = one_iter(measure=True)
+        for k in ("batch", "fwd_bwd", "opt", "ln_ar", "emb_ar", "total"):
+            acc[k] = acc.get(k, 0.0) + m[k]
+        for i, v in enumerate(m["layers"]):
+            layer_acc[i] += v
+    for k in acc:
+        acc[k] /= iters
+    layer_ms = [v / iters for v in layer_acc]
+
+    # per-layer memory: parameter + grad + optimizer state bytes + peak
+    # activation delta measured by the forward hooks
+    def module_state_bytes(module: torch.nn.Module) -> float:
+        n = sum(p.numel() for p in module.parameters())
+        # bf16 param + fp32 grad-flat share + fp32 master + m + v
+        return n * (2 + 4 + 4 + 4 + 4)
+
+    layer_mem_mb = [
+        (module_state_bytes(l) + t.act_bytes) / (1024.0 * 1024.0)
+        for l, t in zip(layers, timers)
+    ]
+
+    for t in timers:
+        t.remove()
+
+    rank = dist.get_rank() if dist.is_initialized() else 0
+    if rank != 0:
+        return None
+    os.makedirs(out_dir, exist_ok=True)
+    path = os.path.join(out_dir, f"DeviceType.{device_type}_tp{tp}_bs{bs}.json")
+    ProfileStore.write_profile_json(
+        path,
+        model_name=spec.name,
+        parameters_per_layer_bytes=model.layer_parameter_bytes(),
+        total_time_ms=acc["total"],
+        forward_backward_time_ms=acc["fwd_bwd"],
+        batch_generator_time_ms=acc["batch"],
+        layernorm_grads_all_reduce_time_ms=acc["ln_ar"],
+        embedding_grads_all_reduce_time_ms=acc["emb_ar"],
+        optimizer_time_ms=acc["opt"],
+        layer_compute_total_ms=layer_ms,
+        total_memory_mb=sum(layer_mem_mb),
+        layer_memory_total_mb=layer_mem_mb,
+    )
+    return path
+
+
+def main() -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--model", default="gpt2-small", choices=sorted(MODEL_SPECS))
+    p.add_argument("--bs", default="1,2,4,8", help="comma-separated batch sizes")
+    p.add_argument("--device-type", default="MI355X")
+    p.add_argument("--out", default="profiles/mi355x")
+    p.add_argument("--iters", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--seq-length", type=int, default=None)
+    args = p.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    tp_group = None
+    if world > 1:
+        from metis_amd.runtime.comm import init_parallel
+
+        ctx = init_parallel(dp=1, tp=world, pp=1)
+        tp_group = ctx.tp_group
+
+    spec = MODEL_SPECS[args.model]
+    for bs in [int(b) for b in args.bs.split(",")]:
+        path = profile_model(
+            spec, bs=bs, tp=world, device_type=args.device_type,
+            out_dir=args.out, warmup=args.warmup, iters=args.iters,
+            tp_group=tp_group, seq_length=args.seq_length,
+        )
+        if path:
+            print(f"wrote {path}")
+
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
