@@ -23,6 +23,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from metis_amd.ops import LayerNorm
+from metis_amd.ops.attention import flash_attention
 
 
 @dataclass(frozen=True)
@@ -198,10 +199,10 @@ class GPTBlock(nn.Module):
         qkv = self.qkv(y, tp_group)
         qkv = qkv.view(b, s, self.heads_per_rank, 3 * self.head_dim)
         q, k, v = qkv.chunk(3, dim=-1)
-        q = q.transpose(1, 2)  # [b, heads, s, d]
-        k = k.transpose(1, 2)
-        v = v.transpose(1, 2)
-        attn = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        q = q.transpose(1, 2).contiguous()  # [b, heads, s, d]
+        k = k.transpose(1, 2).contiguous()
+        v = v.transpose(1, 2).contiguous()
+        attn = flash_attention(q, k, v, causal=True)
         attn = attn.transpose(1, 2).reshape(b, s, -1)
         x = residual + self.proj(attn, tp_group)
 

@@ -1,10 +1,11 @@
-"""Flash attention wrapper (causal, GQA).
+"""Flash attention (causal, GQA) — hand-written gfx950 MFMA kernels.
 
-Forward runs the hand-written gfx950 MFMA kernel; backward currently
-falls back to PyTorch SDPA recompute (the hand-written backward kernel is
-the next optimization stage — until it lands, training graphs that need
-grads route through SDPA, and the custom kernel serves no-grad forward
-passes, decode and the profiler's forward timings).
+Forward: online-softmax flash kernel (attention.hip), saving the row LSE.
+Backward: split dQ / dKV recompute kernels (attention_bwd.hip); GQA dK/dV
+come back per q-head and reduce over the group here (deterministic).
+
+Unsupported shapes (D > 128, D % 16 != 0, S % 64 != 0, non-causal) fall
+back to PyTorch SDPA; on GPU-supported shapes the HIP path always runs.
 """
 
 from __future__ import annotations
@@ -30,6 +31,33 @@ def _supported(q: torch.Tensor) -> bool:
     )
 
 
+class _FlashAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        ext = _ops.require_extension()
+        o, lse = ext.attn_fwd(q, k, v, scale)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, d_o):
+        ext = _ops.require_extension()
+        q, k, v, o, lse = ctx.saved_tensors
+        delta = (d_o.float() * o.float()).sum(-1)   # [B, H, S]
+        dq, dk_h, dv_h = ext.attn_bwd(q, k, v, d_o.contiguous(), lse,
+                                      delta.contiguous(), ctx.scale)
+        h, hkv = q.size(1), k.size(1)
+        if h != hkv:
+            b, _, s, d = q.shape
+            dk = dk_h.view(b, hkv, h // hkv, s, d).sum(2, dtype=torch.float32)
+            dv = dv_h.view(b, hkv, h // hkv, s, d).sum(2, dtype=torch.float32)
+            dk, dv = dk.to(k.dtype), dv.to(v.dtype)
+        else:
+            dk, dv = dk_h, dv_h
+        return dq, dk, dv, None
+
+
 def flash_attention(
     q: torch.Tensor,
     k: torch.Tensor,
@@ -40,13 +68,8 @@ def flash_attention(
     """q [B,H,S,D], k/v [B,Hkv,S,D] -> [B,H,S,D]."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.size(-1))
-    needs_grad = torch.is_grad_enabled() and (
-        q.requires_grad or k.requires_grad or v.requires_grad
-    )
-    if causal and not needs_grad and _supported(q):
-        ext = _ops.require_extension()
-        o, _lse = ext.attn_fwd(q, k, v, scale)
-        return o
+    if causal and _supported(q):
+        return _FlashAttention.apply(q, k, v, scale)
     return F.scaled_dot_product_attention(
         q, k, v, is_causal=causal, scale=scale, enable_gqa=q.size(1) != k.size(1)
     )

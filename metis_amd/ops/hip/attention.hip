@@ -103,6 +103,14 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
         }
         __syncthreads();
 
+        // Causal: this wave's rows end at q0+15; a tile starting past that
+        // is fully masked — with m_run still at -1e30 its exp(s - m) would
+        // be exp(0) = 1, so skip the compute (barriers stay block-wide).
+        if (kv0 > q0 + 15) {
+            __syncthreads();
+            continue;
+        }
+
         // ---- S = scale * Q @ K^T for two 16-col subtiles ----------------
         floatx4 s_acc[2];
         #pragma unroll

@@ -15,6 +15,10 @@ void adamw_step(
     double weight_decay, long step, double grad_scale);
 std::vector<torch::Tensor> attn_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v, double scale);
+std::vector<torch::Tensor> attn_bwd(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor d_o, torch::Tensor lse, torch::Tensor delta,
+    double scale);
 torch::Tensor gemm_bf16(
     torch::Tensor a, torch::Tensor b_nk, c10::optional<torch::Tensor> bias,
     long epilogue);
@@ -28,6 +32,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("adamw_step", &adamw_step, "fused AdamW step");
     m.def("attn_fwd", &attn_fwd,
           "flash attention forward (causal, GQA) -> (O, LSE)");
+    m.def("attn_bwd", &attn_bwd,
+          "flash attention backward -> (dQ, dK_per_head, dV_per_head)");
     m.def("gemm_bf16", &gemm_bf16,
           "MFMA bf16 GEMM A[M,K] @ B[N,K]^T with fused epilogue "
           "(0=none, 1=bias, 2=bias+gelu)");

@@ -22,6 +22,7 @@ sources = [
     os.path.join(HIP_DIR, "layernorm.hip"),
     os.path.join(HIP_DIR, "adamw.hip"),
     os.path.join(HIP_DIR, "attention.hip"),
+    os.path.join(HIP_DIR, "attention_bwd.hip"),
     os.path.join(HIP_DIR, "gemm.hip"),
 ]
 sources = [s for s in sources if os.path.exists(s)]
