@@ -1,0 +1,174 @@
+"""Llama model family: RMSNorm + RoPE + GQA flash attention + SwiGLU MLP,
+with the same Megatron-style TP sharding and pipeline-stage slicing as
+the GPT family (profile layer convention: 0 = embed, 1..n-2 = blocks,
+n-1 = final norm + head)."""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+import torch.nn.functional as F
+
+from metis_amd.models.gpt import (
+    ColumnParallelLinear,
+    RowParallelLinear,
+    _VocabParallelCrossEntropy,
+)
+from metis_amd.ops.attention import flash_attention
+from metis_amd.ops.norms import RMSNorm, apply_rope, swiglu
+
+
+@dataclass(frozen=True)
+class LlamaModelSpec:
+    name: str
+    hidden_size: int
+    num_layers: int
+    num_heads: int
+    num_kv_heads: int
+    ffn_hidden_size: int
+    vocab_size: int
+    seq_length: int
+    rope_base: float = 500000.0
+
+    @property
+    def head_dim(self) -> int:
+        return self.hidden_size // self.num_heads
+
+    @property
+    def profile_num_layers(self) -> int:
+        return self.num_layers + 2
+
+    def num_parameters(self) -> int:
+        h, v, f = self.hidden_size, self.vocab_size, self.ffn_hidden_size
+        kvh = self.num_kv_heads * self.head_dim
+        per_block = h * h + 2 * h * kvh + h * h + 3 * h * f + 2 * h
+        return 2 * v * h + self.num_layers * per_block + h
+
+
+LLAMA_SPECS = {
+    # Llama-3-8B shape (vocab rounded to a 128-multiple for TP sharding)
+    "llama3-8b": LlamaModelSpec("llama3-8b", 4096, 32, 32, 8, 14336,
+                                128256, 8192),
+    "llama3-1b": LlamaModelSpec("llama3-1b", 2048, 16, 32, 8, 8192,
+                                128256, 2048),
+    "llama-tiny": LlamaModelSpec("llama-tiny", 256, 2, 4, 2, 512, 2048, 256),
+}
+
+
+class LlamaBlock(nn.Module):
+    def __init__(self, spec: LlamaModelSpec, tp: int, dtype):
+        super().__init__()
+        h = spec.hidden_size
+        d = spec.head_dim
+        assert spec.num_heads % tp == 0 and spec.num_kv_heads % tp == 0, (
+            "heads and kv heads must divide by tp"
+        )
+        self.heads_per_rank = spec.num_heads // tp
+        self.kv_heads_per_rank = spec.num_kv_heads // tp
+        self.head_dim = d
+        self.rope_base = spec.rope_base
+
+        qkv_out = (spec.num_heads + 2 * spec.num_kv_heads) * d
+        self.norm_attn = RMSNorm(h)
+        self.qkv = ColumnParallelLinear(h, qkv_out, tp, dtype)
+        self.proj = RowParallelLinear(spec.num_heads * d, h, tp, dtype)
+        self.norm_mlp = RMSNorm(h)
+        # gate & up fused in one column-parallel GEMM
+        self.gate_up = ColumnParallelLinear(h, 2 * spec.ffn_hidden_size, tp, dtype)
+        self.down = RowParallelLinear(spec.ffn_hidden_size, h, tp, dtype)
+        self.ffn_per_rank = spec.ffn_hidden_size // tp
+
+    def forward(self, x: torch.Tensor, tp_group) -> torch.Tensor:
+        b, s, _ = x.shape
+        hq, hkv, d = self.heads_per_rank, self.kv_heads_per_rank, self.head_dim
+
+        residual = x
+        y = self.norm_attn(x)
+        qkv = self.qkv(y, tp_group)
+        q, k, v = qkv.split(
+            [hq * d, hkv * d, hkv * d], dim=-1
+        )
+        q = q.view(b, s, hq, d).transpose(1, 2).contiguous()
+        k = k.view(b, s, hkv, d).transpose(1, 2).contiguous()
+        v = v.view(b, s, hkv, d).transpose(1, 2).contiguous()
+        q = apply_rope(q, self.rope_base)
+        k = apply_rope(k, self.rope_base)
+        attn = flash_attention(q, k, v, causal=True)
+        attn = attn.transpose(1, 2).reshape(b, s, hq * d)
+        x = residual + self.proj(attn, tp_group)
+
+        residual = x
+        y = self.norm_mlp(x)
+        gate_up = self.gate_up(y, tp_group)
+        gate, up = gate_up.split([self.ffn_per_rank, self.ffn_per_rank], dim=-1)
+        x = residual + self.down(swiglu(gate.contiguous(), up.contiguous()),
+                                 tp_group)
+        return x
+
+
+class LlamaModel(nn.Module):
+    """A pipeline-stage slice of the Llama model (see GPTModel docs)."""
+
+    def __init__(
+        self,
+        spec: LlamaModelSpec,
+        tp: int = 1,
+        dtype: torch.dtype = torch.bfloat16,
+        layer_range: Optional[tuple] = None,
+        tp_group=None,
+    ):
+        super().__init__()
+        self.spec = spec
+        self.tp = tp
+        self.tp_group = tp_group
+        total = spec.profile_num_layers
+        start, end = layer_range if layer_range is not None else (0, total)
+        self.has_embedding = start == 0
+        self.has_head = end == total
+
+        h = spec.hidden_size
+        if self.has_embedding:
+            self.wte = nn.Embedding(spec.vocab_size, h, dtype=dtype)
+            nn.init.normal_(self.wte.weight, std=0.02)
+
+        block_start = max(start - 1, 0)
+        block_end = min(end, total - 1) - 1
+        self.blocks = nn.ModuleList(
+            LlamaBlock(spec, tp, dtype)
+            for _ in range(max(block_end - block_start, 0))
+        )
+        if self.has_head:
+            self.norm_final = RMSNorm(h)
+            self.head = ColumnParallelLinear(h, spec.vocab_size, tp, dtype)
+
+    def forward(self, x, labels=None):
+        if self.has_embedding:
+            x = self.wte(x)
+        for block in self.blocks:
+            x = block(x, self.tp_group)
+        if self.has_head:
+            x = self.norm_final(x)
+            logits = self.head(x, self.tp_group)
+            if labels is not None:
+                flat = logits.float().view(-1, logits.size(-1))
+                labels = labels.reshape(-1)
+                if self.tp_group is not None and dist.get_world_size(self.tp_group) > 1:
+                    return _VocabParallelCrossEntropy.apply(flat, labels, self.tp_group)
+                return F.cross_entropy(flat, labels)
+            return logits
+        return x
+
+    def layer_parameter_bytes(self) -> List[float]:
+        spec = self.spec
+        el = 2
+        h, v, f = spec.hidden_size, spec.vocab_size, spec.ffn_hidden_size
+        d = spec.head_dim
+        kvh = spec.num_kv_heads * d
+        embed = v * h * el
+        per_block = (h * h + 2 * h * kvh + h * h + 3 * h * f + 2 * h) * el
+        head = (v * h + h) * el
+        return [float(embed)] + [float(per_block)] * spec.num_layers + [float(head)]

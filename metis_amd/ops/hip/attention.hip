@@ -103,15 +103,15 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
         }
         __syncthreads();
 
-        // Causal: this wave's rows end at q0+15; a tile starting past that
-        // is fully masked — with m_run still at -1e30 its exp(s - m) would
-        // be exp(0) = 1, so skip the compute (barriers stay block-wide).
-        if (kv0 > q0 + 15) {
-            __syncthreads();
-            continue;
-        }
+        // Causal: a tile starting past this wave's last row (q0+15) is
+        // fully masked — with m_run still at -1e30 its exp(s - m) would be
+        // exp(0) = 1. Inactive waves skip the compute but keep every
+        // barrier (uniform control flow).
+        const bool active = kv0 <= q0 + 15;
 
         // ---- S = scale * Q @ K^T for two 16-col subtiles ----------------
+        float p[2][4];   // exp(S - m) per (j, r)
+        if (active) {
         floatx4 s_acc[2];
         #pragma unroll
         for (int j = 0; j < 2; ++j) {
@@ -133,7 +133,6 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
         }
 
         // ---- causal mask + online softmax -------------------------------
-        float p[2][4];   // exp(S - m) per (j, r)
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
             const int qrow = q0 + k8 * 4 + r;
@@ -170,24 +169,32 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
                 o_acc[jd][r] *= alpha;
         }
 
-        // ---- P (C layout) -> LDS -> A-layout fragments ------------------
+        // ---- P (C layout) -> LDS ----------------------------------------
         #pragma unroll
         for (int j = 0; j < 2; ++j)
             #pragma unroll
             for (int r = 0; r < 4; ++r)
                 Pw[(k8 * 4 + r) * KVBLK + j * 16 + col16] =
                     __float2bfloat16(p[j][r]);
-        // wave-local LDS: no barrier needed (single-wave producer/consumer)
+        }  // active
+
+        // The barrier both keeps control flow uniform and orders the
+        // scalar P stores against the vector re-read below: the two go
+        // through different pointer types, and without a barrier the
+        // compiler may hoist the ds_read above the ds_writes (TBAA).
+        __syncthreads();
+
+        if (active) {
+        // ---- O += P @ V (P re-read in A layout) -------------------------
         bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(
             Pw + col16 * KVBLK + k8 * 8);
-
-        // ---- O += P @ V -------------------------------------------------
         for (int jd = 0; jd < djtiles; ++jd) {
             bf16x8 v_frag = *reinterpret_cast<const bf16x8*>(
                 Vt + (jd * 16 + col16) * KVBLK + k8 * 8);
             o_acc[jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 p_frag, v_frag, o_acc[jd], 0, 0, 0);
         }
+        }  // active
         __syncthreads();   // Vt will be overwritten next tile
     }
 

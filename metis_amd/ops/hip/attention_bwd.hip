@@ -104,11 +104,10 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
             }
         }
         __syncthreads();
-        if (kv0 > q0 + 15) {   // fully masked for this wave
-            __syncthreads();
-            continue;
-        }
+        // inactive waves skip compute but keep barriers uniform
+        const bool active = kv0 <= q0 + 15;
 
+        if (active) {
         // S and dP for two 16-col subtiles
         floatx4 s_acc[2], dp_acc[2];
         #pragma unroll
@@ -140,7 +139,13 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
                 float ds = scale * p * (dp_acc[j][r] - delta_r[r]);
                 Sw[(k8 * 4 + r) * CTILE + j * 16 + col16] = __float2bfloat16(ds);
             }
+        }  // active
 
+        // barrier orders the scalar dS stores against the vector re-read
+        // (different pointer types: TBAA would otherwise allow hoisting)
+        __syncthreads();
+
+        if (active) {
         // dQ += dS @ K : A = dS (LDS relayout), B = K^T from LDS
         bf16x8 ds_frag = *reinterpret_cast<const bf16x8*>(
             Sw + col16 * CTILE + k8 * 8);
@@ -150,6 +155,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
             dq_acc[jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 ds_frag, kt_frag, dq_acc[jd], 0, 0, 0);
         }
+        }  // active
         __syncthreads();
     }
 
@@ -192,6 +198,8 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     bf16* dOt = reinterpret_cast<bf16*>(smem + DMAX * CTILE * 2);    // [D][32]
     bf16* Sw = reinterpret_cast<bf16*>(smem + 2 * DMAX * CTILE * 2)
                + wave * 16 * CTILE;                                   // [16][32]
+    bf16* Pw = reinterpret_cast<bf16*>(smem + 2 * DMAX * CTILE * 2
+               + 4 * 16 * CTILE * 2) + wave * 16 * CTILE;             // [16][32]
 
     // K and V fragments for this wave's rows (A layout, m = col16)
     bf16x8 k_frag[4], v_frag[4];
@@ -229,11 +237,11 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
             }
         }
         __syncthreads();
-        if (q0 + CTILE - 1 < kv0) {   // entire q tile above the diagonal
-            __syncthreads();
-            continue;
-        }
+        // entire q tile above the diagonal -> inactive (barriers uniform)
+        const bool active = q0 + CTILE - 1 >= kv0;
 
+        float pt[2][4];
+        if (active) {
         // S^T = K Q^T and dP^T = V dO^T for two 16-col (q) subtiles
         floatx4 st_acc[2], dpt_acc[2];
         #pragma unroll
@@ -253,7 +261,6 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
         }
 
         // P^T and dS^T (C layout: row = kv = k8*4+r, col = q)
-        float pt[2][4];
         #pragma unroll
         for (int j = 0; j < 2; ++j)
             #pragma unroll
@@ -268,8 +275,16 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
                 float ds = scale * p
                            * (dpt_acc[j][r] - Delta[row_base + qcol]);
                 Sw[(k8 * 4 + r) * CTILE + j * 16 + col16] = __float2bfloat16(ds);
+                Pw[(k8 * 4 + r) * CTILE + j * 16 + col16] =
+                    __float2bfloat16(pt[j][r]);
             }
+        }  // active
 
+        // barrier orders the scalar dS^T / P^T stores against the vector
+        // re-reads (TBAA) and keeps control flow uniform
+        __syncthreads();
+
+        if (active) {
         // dK += dS^T @ Q : A = dS^T (LDS relayout), B = Q^T (LDS)
         {
             bf16x8 dst_frag = *reinterpret_cast<const bf16x8*>(
@@ -283,15 +298,9 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
         }
 
         // dV += P^T @ dO : A = P^T (LDS relayout), B = dO^T (LDS)
-        #pragma unroll
-        for (int j = 0; j < 2; ++j)
-            #pragma unroll
-            for (int r = 0; r < 4; ++r)
-                Sw[(k8 * 4 + r) * CTILE + j * 16 + col16] =
-                    __float2bfloat16(pt[j][r]);
         {
             bf16x8 pt_frag = *reinterpret_cast<const bf16x8*>(
-                Sw + col16 * CTILE + k8 * 8);
+                Pw + col16 * CTILE + k8 * 8);
             for (int jd = 0; jd < djtiles; ++jd) {
                 bf16x8 dot_frag = *reinterpret_cast<const bf16x8*>(
                     dOt + (jd * 16 + col16) * CTILE + k8 * 8);
@@ -299,6 +308,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
                     pt_frag, dot_frag, dv_acc[jd], 0, 0, 0);
             }
         }
+        }  // active
         __syncthreads();
     }
 
@@ -336,7 +346,7 @@ std::vector<torch::Tensor> attn_bwd(
     auto stream = c10::hip::getCurrentHIPStream().stream();
     const int grid = (int)(B * H * (S / RBLK));
     const int lds_dq = DMAX * CTILE * 2 + 4 * 16 * CTILE * 2;
-    const int lds_dkv = 2 * DMAX * CTILE * 2 + 4 * 16 * CTILE * 2;
+    const int lds_dkv = 2 * DMAX * CTILE * 2 + 2 * 4 * 16 * CTILE * 2;
 
     hipLaunchKernelGGL(attn_bwd_dq_kernel, dim3(grid), dim3(THREADS), lds_dq,
         stream,

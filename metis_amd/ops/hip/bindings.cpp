@@ -23,6 +23,15 @@ torch::Tensor gemm_bf16(
     torch::Tensor a, torch::Tensor b_nk, c10::optional<torch::Tensor> bias,
     long epilogue);
 torch::Tensor mfma_tile_probe(torch::Tensor a, torch::Tensor b);
+std::vector<torch::Tensor> rmsnorm_fwd(
+    torch::Tensor x, torch::Tensor gamma, double eps);
+std::vector<torch::Tensor> rmsnorm_bwd(
+    torch::Tensor dy, torch::Tensor x, torch::Tensor gamma, torch::Tensor rstd);
+torch::Tensor rope_apply(
+    torch::Tensor x, torch::Tensor cos_t, torch::Tensor sin_t, bool backward);
+torch::Tensor swiglu_fwd(torch::Tensor a, torch::Tensor b);
+std::vector<torch::Tensor> swiglu_bwd(
+    torch::Tensor dy, torch::Tensor a, torch::Tensor b);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("layernorm_fwd", &layernorm_fwd,
@@ -39,4 +48,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "(0=none, 1=bias, 2=bias+gelu)");
     m.def("mfma_tile_probe", &mfma_tile_probe,
           "single-wave 16x16x32 MFMA layout probe");
+    m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused RMSNorm forward");
+    m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused RMSNorm backward");
+    m.def("rope_apply", &rope_apply, "rotary embedding (fwd/bwd by flag)");
+    m.def("swiglu_fwd", &swiglu_fwd, "fused silu(a)*b");
+    m.def("swiglu_bwd", &swiglu_bwd, "fused SwiGLU backward");
 }
