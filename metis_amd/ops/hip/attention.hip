@@ -29,13 +29,14 @@ constexpr int QBLK = 64;      // q rows per block (16 per wave)
 constexpr int KVBLK = 32;     // kv columns per tile
 constexpr int DMAX = 128;
 
+template <int D>
 __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
     const bf16* __restrict__ Q,   // [B, H, S, D]
     const bf16* __restrict__ K,   // [B, Hkv, S, D]
     const bf16* __restrict__ V,   // [B, Hkv, S, D]
     bf16* __restrict__ O,         // [B, H, S, D]
     float* __restrict__ LSE,      // [B, H, S]
-    int B, int H, int Hkv, int S, int D,
+    int B, int H, int Hkv, int S,
     float scale) {
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
@@ -51,21 +52,25 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
     const long kv_base = (((long)batch * Hkv + kv_head) * S) * D;
 
     const int q0 = qtile * QBLK + wave * 16;   // this wave's first q row
-    const int dchunks = (D + 31) / 32;         // 32-wide D chunks
-    const int djtiles = D / 16;                // 16-wide output column tiles
+    constexpr int dchunks = (D + 31) / 32;     // 32-wide D chunks
+    constexpr int djtiles = D / 16;            // 16-wide output column tiles
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    bf16* Vt = reinterpret_cast<bf16*>(smem);               // [D][KVBLK]
-    bf16* Pw = reinterpret_cast<bf16*>(smem + DMAX * KVBLK * 2)
-               + wave * 16 * KVBLK;                          // per-wave [16][KVBLK]
+    // LDS tiles hold raw bf16 BITS as shorts: assigning a short into a
+    // __hip_bfloat16 struct would numerically convert (16256 -> 16256.0f),
+    // not reinterpret — the classic trap this kernel once had.
+    short* Vt = reinterpret_cast<short*>(smem);             // [D][KVBLK]
+    short* Pw = reinterpret_cast<short*>(smem + DMAX * KVBLK * 2)
+                + wave * 16 * KVBLK;                         // per-wave [16][KVBLK]
 
     // ---- preload this wave's Q fragments (A-layout per 32-chunk) --------
     bf16x8 q_frag[4];
     {
         const int qrow = q0 + col16;
-        for (int c = 0; c < 4; ++c) {
+        #pragma unroll
+        for (int c = 0; c < dchunks; ++c) {
             const int d0 = c * 32 + k8 * 8;
-            if (c < dchunks && d0 < D) {
+            if (d0 < D) {
                 q_frag[c] = *reinterpret_cast<const bf16x8*>(
                     Q + q_base + (long)qrow * D + d0);
             } else {
@@ -82,7 +87,8 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
         m_run[r] = -1e30f;
         l_run[r] = 0.f;
     }
-    floatx4 o_acc[DMAX / 16];
+    floatx4 o_acc[djtiles];
+    #pragma unroll
     for (int jd = 0; jd < djtiles; ++jd) o_acc[jd] = floatx4{0.f, 0.f, 0.f, 0.f};
 
     const int kv_end = qtile * QBLK + QBLK;    // causal bound for the block
@@ -90,7 +96,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
         // ---- stage V^T into LDS (whole block cooperates) ----------------
         {
             // 256 threads x bf16x8: covers KVBLK * D / 8 chunks
-            const int chunks = KVBLK * D / 8;
+            constexpr int chunks = KVBLK * D / 8;
             for (int c = threadIdx.x; c < chunks; c += THREADS) {
                 const int row = c / (D / 8);        // kv row in tile
                 const int d0 = (c % (D / 8)) * 8;
@@ -117,6 +123,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
         for (int j = 0; j < 2; ++j) {
             s_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
             const int kvrow = kv0 + j * 16 + col16;
+            #pragma unroll
             for (int c = 0; c < dchunks; ++c) {
                 bf16x8 k_frag;
                 const int d0 = c * 32 + k8 * 8;
@@ -165,6 +172,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
             l_run[r] = l_run[r] * alpha + row_sum;
             m_run[r] = m_new;
             // rescale O rows r
+            #pragma unroll
             for (int jd = 0; jd < djtiles; ++jd)
                 o_acc[jd][r] *= alpha;
         }
@@ -175,7 +183,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
             #pragma unroll
             for (int r = 0; r < 4; ++r)
                 Pw[(k8 * 4 + r) * KVBLK + j * 16 + col16] =
-                    __float2bfloat16(p[j][r]);
+                    float_to_bf16_bits(p[j][r]);
         }  // active
 
         // The barrier both keeps control flow uniform and orders the
@@ -188,6 +196,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
         // ---- O += P @ V (P re-read in A layout) -------------------------
         bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(
             Pw + col16 * KVBLK + k8 * 8);
+        #pragma unroll
         for (int jd = 0; jd < djtiles; ++jd) {
             bf16x8 v_frag = *reinterpret_cast<const bf16x8*>(
                 Vt + (jd * 16 + col16) * KVBLK + k8 * 8);
@@ -203,6 +212,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
     for (int r = 0; r < 4; ++r) {
         const int qrow = q0 + k8 * 4 + r;
         const float inv_l = 1.f / l_run[r];
+        #pragma unroll
         for (int jd = 0; jd < djtiles; ++jd)
             O[q_base + (long)qrow * D + jd * 16 + col16] =
                 __float2bfloat16(o_acc[jd][r] * inv_l);
@@ -230,15 +240,24 @@ std::vector<torch::Tensor> attn_fwd(
 
     const int grid = (int)(B * H * (S / QBLK));
     const int lds = DMAX * KVBLK * 2 + 4 * 16 * KVBLK * 2;
-    hipLaunchKernelGGL(
-        attn_fwd_kernel, dim3(grid), dim3(THREADS), lds,
-        c10::hip::getCurrentHIPStream().stream(),
-        reinterpret_cast<const bf16*>(qc.data_ptr()),
-        reinterpret_cast<const bf16*>(kc.data_ptr()),
-        reinterpret_cast<const bf16*>(vc.data_ptr()),
-        reinterpret_cast<bf16*>(o.data_ptr()),
-        lse.data_ptr<float>(),
-        (int)B, (int)H, (int)Hkv, (int)S, (int)D, (float)scale);
+    auto stream = c10::hip::getCurrentHIPStream().stream();
+    #define LAUNCH_D(DD)                                                     \
+        hipLaunchKernelGGL(attn_fwd_kernel<DD>, dim3(grid), dim3(THREADS),   \
+            lds, stream,                                                     \
+            reinterpret_cast<const bf16*>(qc.data_ptr()),                    \
+            reinterpret_cast<const bf16*>(kc.data_ptr()),                    \
+            reinterpret_cast<const bf16*>(vc.data_ptr()),                    \
+            reinterpret_cast<bf16*>(o.data_ptr()),                           \
+            lse.data_ptr<float>(),                                           \
+            (int)B, (int)H, (int)Hkv, (int)S, (float)scale)
+    switch (D) {
+        case 64: LAUNCH_D(64); break;
+        case 80: LAUNCH_D(80); break;
+        case 96: LAUNCH_D(96); break;
+        case 128: LAUNCH_D(128); break;
+        default: TORCH_CHECK(false, "head dim must be 64/80/96/128, got ", D);
+    }
+    #undef LAUNCH_D
     HIP_CHECK_LAST();
     return {o, lse};
 }

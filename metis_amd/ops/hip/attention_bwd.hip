@@ -29,22 +29,23 @@ constexpr int CTILE = 32;    // opposing-side tile width
 constexpr int DMAX = 128;
 
 // load one natural A/B fragment: 8 contiguous bf16 of `row`, chunk c
-__device__ __forceinline__ bf16x8 frag8(const bf16* base, long row, int D,
-                                        int d0) {
+template <int D>
+__device__ __forceinline__ bf16x8 frag8(const bf16* base, long row, int d0) {
     if (d0 < D)
         return *reinterpret_cast<const bf16x8*>(base + row * D + d0);
     bf16x8 z;
     #pragma unroll
-    for (int i = 0; i < 8; ++i) z[i] = bf16(0.f);
+    for (int i = 0; i < 8; ++i) z[i] = 0;
     return z;
 }
 
+template <int D>
 __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, const bf16* __restrict__ dO,
     const float* __restrict__ LSE, const float* __restrict__ Delta,
     bf16* __restrict__ dQ,
-    int B, int H, int Hkv, int S, int D, float scale) {
+    int B, int H, int Hkv, int S, float scale) {
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
     const int col16 = lane & 15;
@@ -59,13 +60,13 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
     const long row_base = ((long)batch * H + head) * S;
 
     const int q0 = qtile * RBLK + wave * 16;
-    const int dchunks = (D + 31) / 32;
-    const int djtiles = D / 16;
+    constexpr int dchunks = (D + 31) / 32;
+    constexpr int djtiles = D / 16;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    bf16* Kt = reinterpret_cast<bf16*>(smem);                       // [D][32]
-    bf16* Sw = reinterpret_cast<bf16*>(smem + DMAX * CTILE * 2)
-               + wave * 16 * CTILE;                                  // [16][32]
+    short* Kt = reinterpret_cast<short*>(smem);                     // [D][32]
+    short* Sw = reinterpret_cast<short*>(smem + DMAX * CTILE * 2)
+                + wave * 16 * CTILE;                                 // [16][32]
 
     // per-lane row state (rows k8*4 + r)
     float lse_r[4], delta_r[4];
@@ -77,22 +78,22 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
     }
 
     // Q and dO fragments for this wave's rows (A layout, m = col16)
-    bf16x8 q_frag[4], do_frag[4];
-    for (int c = 0; c < 4; ++c) {
-        q_frag[c] = frag8(Q + q_base, q0 + col16, D,
-                          c < dchunks ? c * 32 + k8 * 8 : DMAX + 8);
-        do_frag[c] = frag8(dO + q_base, q0 + col16, D,
-                           c < dchunks ? c * 32 + k8 * 8 : DMAX + 8);
+    bf16x8 q_frag[dchunks], do_frag[dchunks];
+    #pragma unroll
+    for (int c = 0; c < dchunks; ++c) {
+        q_frag[c] = frag8<D>(Q + q_base, q0 + col16, c * 32 + k8 * 8);
+        do_frag[c] = frag8<D>(dO + q_base, q0 + col16, c * 32 + k8 * 8);
     }
 
-    floatx4 dq_acc[DMAX / 16];
+    floatx4 dq_acc[djtiles];
+    #pragma unroll
     for (int jd = 0; jd < djtiles; ++jd) dq_acc[jd] = floatx4{0.f, 0.f, 0.f, 0.f};
 
     const int kv_end = qtile * RBLK + RBLK;
     for (int kv0 = 0; kv0 < kv_end; kv0 += CTILE) {
         // stage K^T tile into LDS (block-wide)
         {
-            const int chunks = CTILE * D / 8;
+            constexpr int chunks = CTILE * D / 8;
             for (int c = threadIdx.x; c < chunks; c += THREADS) {
                 const int row = c / (D / 8);
                 const int d0 = (c % (D / 8)) * 8;
@@ -115,12 +116,13 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
             s_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
             dp_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
             const long kvrow = kv0 + j * 16 + col16;
+            #pragma unroll
             for (int c = 0; c < dchunks; ++c) {
                 const int d0 = c * 32 + k8 * 8;
-                bf16x8 k_frag = frag8(K + kv_base, kvrow, D, d0);
+                bf16x8 k_frag = frag8<D>(K + kv_base, kvrow, d0);
                 s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     q_frag[c], k_frag, s_acc[j], 0, 0, 0);
-                bf16x8 v_frag = frag8(V + kv_base, kvrow, D, d0);
+                bf16x8 v_frag = frag8<D>(V + kv_base, kvrow, d0);
                 dp_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     do_frag[c], v_frag, dp_acc[j], 0, 0, 0);
             }
@@ -137,7 +139,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
                               ? 0.f
                               : __expf(s_acc[j][r] * scale - lse_r[r]);
                 float ds = scale * p * (dp_acc[j][r] - delta_r[r]);
-                Sw[(k8 * 4 + r) * CTILE + j * 16 + col16] = __float2bfloat16(ds);
+                Sw[(k8 * 4 + r) * CTILE + j * 16 + col16] = float_to_bf16_bits(ds);
             }
         }  // active
 
@@ -149,6 +151,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
         // dQ += dS @ K : A = dS (LDS relayout), B = K^T from LDS
         bf16x8 ds_frag = *reinterpret_cast<const bf16x8*>(
             Sw + col16 * CTILE + k8 * 8);
+        #pragma unroll
         for (int jd = 0; jd < djtiles; ++jd) {
             bf16x8 kt_frag = *reinterpret_cast<const bf16x8*>(
                 Kt + (jd * 16 + col16) * CTILE + k8 * 8);
@@ -162,19 +165,21 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
         const long qrow = q0 + k8 * 4 + r;
+        #pragma unroll
         for (int jd = 0; jd < djtiles; ++jd)
             dQ[q_base + qrow * D + jd * 16 + col16] =
                 __float2bfloat16(dq_acc[jd][r]);
     }
 }
 
+template <int D>
 __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, const bf16* __restrict__ dO,
     const float* __restrict__ LSE, const float* __restrict__ Delta,
     bf16* __restrict__ dK,   // [B, H, S, D] per q-head (wrapper reduces GQA)
     bf16* __restrict__ dV,
-    int B, int H, int Hkv, int S, int D, float scale) {
+    int B, int H, int Hkv, int S, float scale) {
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
     const int col16 = lane & 15;
@@ -190,27 +195,27 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     const long row_base = ((long)batch * H + head) * S;
 
     const int kv0 = kvtile * RBLK + wave * 16;   // wave's first kv row
-    const int dchunks = (D + 31) / 32;
-    const int djtiles = D / 16;
+    constexpr int dchunks = (D + 31) / 32;
+    constexpr int djtiles = D / 16;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    bf16* Qt = reinterpret_cast<bf16*>(smem);                        // [D][32]
-    bf16* dOt = reinterpret_cast<bf16*>(smem + DMAX * CTILE * 2);    // [D][32]
-    bf16* Sw = reinterpret_cast<bf16*>(smem + 2 * DMAX * CTILE * 2)
-               + wave * 16 * CTILE;                                   // [16][32]
-    bf16* Pw = reinterpret_cast<bf16*>(smem + 2 * DMAX * CTILE * 2
-               + 4 * 16 * CTILE * 2) + wave * 16 * CTILE;             // [16][32]
+    short* Qt = reinterpret_cast<short*>(smem);                      // [D][32]
+    short* dOt = reinterpret_cast<short*>(smem + DMAX * CTILE * 2);   // [D][32]
+    short* Sw = reinterpret_cast<short*>(smem + 2 * DMAX * CTILE * 2)
+                + wave * 16 * CTILE;                                  // [16][32]
+    short* Pw = reinterpret_cast<short*>(smem + 2 * DMAX * CTILE * 2
+                + 4 * 16 * CTILE * 2) + wave * 16 * CTILE;            // [16][32]
 
     // K and V fragments for this wave's rows (A layout, m = col16)
-    bf16x8 k_frag[4], v_frag[4];
-    for (int c = 0; c < 4; ++c) {
-        k_frag[c] = frag8(K + kv_base, kv0 + col16, D,
-                          c < dchunks ? c * 32 + k8 * 8 : DMAX + 8);
-        v_frag[c] = frag8(V + kv_base, kv0 + col16, D,
-                          c < dchunks ? c * 32 + k8 * 8 : DMAX + 8);
+    bf16x8 k_frag[dchunks], v_frag[dchunks];
+    #pragma unroll
+    for (int c = 0; c < dchunks; ++c) {
+        k_frag[c] = frag8<D>(K + kv_base, kv0 + col16, c * 32 + k8 * 8);
+        v_frag[c] = frag8<D>(V + kv_base, kv0 + col16, c * 32 + k8 * 8);
     }
 
-    floatx4 dk_acc[DMAX / 16], dv_acc[DMAX / 16];
+    floatx4 dk_acc[djtiles], dv_acc[djtiles];
+    #pragma unroll
     for (int jd = 0; jd < djtiles; ++jd) {
         dk_acc[jd] = floatx4{0.f, 0.f, 0.f, 0.f};
         dv_acc[jd] = floatx4{0.f, 0.f, 0.f, 0.f};
@@ -221,7 +226,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     for (int q0 = q_start; q0 < S; q0 += CTILE) {
         // stage Q^T and dO^T tiles (block-wide)
         {
-            const int chunks = CTILE * D / 8;
+            constexpr int chunks = CTILE * D / 8;
             for (int c = threadIdx.x; c < chunks; c += THREADS) {
                 const int row = c / (D / 8);
                 const int d0 = (c % (D / 8)) * 8;
@@ -249,12 +254,13 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
             st_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
             dpt_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
             const long qrow = q0 + j * 16 + col16;
+            #pragma unroll
             for (int c = 0; c < dchunks; ++c) {
                 const int d0 = c * 32 + k8 * 8;
-                bf16x8 qf = frag8(Q + q_base, qrow, D, d0);
+                bf16x8 qf = frag8<D>(Q + q_base, qrow, d0);
                 st_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     k_frag[c], qf, st_acc[j], 0, 0, 0);
-                bf16x8 dof = frag8(dO + q_base, qrow, D, d0);
+                bf16x8 dof = frag8<D>(dO + q_base, qrow, d0);
                 dpt_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     v_frag[c], dof, dpt_acc[j], 0, 0, 0);
             }
@@ -274,9 +280,9 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
                 pt[j][r] = p;
                 float ds = scale * p
                            * (dpt_acc[j][r] - Delta[row_base + qcol]);
-                Sw[(k8 * 4 + r) * CTILE + j * 16 + col16] = __float2bfloat16(ds);
+                Sw[(k8 * 4 + r) * CTILE + j * 16 + col16] = float_to_bf16_bits(ds);
                 Pw[(k8 * 4 + r) * CTILE + j * 16 + col16] =
-                    __float2bfloat16(pt[j][r]);
+                    float_to_bf16_bits(pt[j][r]);
             }
         }  // active
 
@@ -289,6 +295,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
         {
             bf16x8 dst_frag = *reinterpret_cast<const bf16x8*>(
                 Sw + col16 * CTILE + k8 * 8);
+            #pragma unroll
             for (int jd = 0; jd < djtiles; ++jd) {
                 bf16x8 qt_frag = *reinterpret_cast<const bf16x8*>(
                     Qt + (jd * 16 + col16) * CTILE + k8 * 8);
@@ -301,6 +308,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
         {
             bf16x8 pt_frag = *reinterpret_cast<const bf16x8*>(
                 Pw + col16 * CTILE + k8 * 8);
+            #pragma unroll
             for (int jd = 0; jd < djtiles; ++jd) {
                 bf16x8 dot_frag = *reinterpret_cast<const bf16x8*>(
                     dOt + (jd * 16 + col16) * CTILE + k8 * 8);
@@ -315,6 +323,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
         const long kvrow = kv0 + k8 * 4 + r;
+        #pragma unroll
         for (int jd = 0; jd < djtiles; ++jd) {
             dK[out_base + kvrow * D + jd * 16 + col16] =
                 __float2bfloat16(dk_acc[jd][r]);
@@ -348,27 +357,35 @@ std::vector<torch::Tensor> attn_bwd(
     const int lds_dq = DMAX * CTILE * 2 + 4 * 16 * CTILE * 2;
     const int lds_dkv = 2 * DMAX * CTILE * 2 + 2 * 4 * 16 * CTILE * 2;
 
-    hipLaunchKernelGGL(attn_bwd_dq_kernel, dim3(grid), dim3(THREADS), lds_dq,
-        stream,
-        reinterpret_cast<const bf16*>(qc.data_ptr()),
-        reinterpret_cast<const bf16*>(kc.data_ptr()),
-        reinterpret_cast<const bf16*>(vc.data_ptr()),
-        reinterpret_cast<const bf16*>(doc.data_ptr()),
-        lsec.data_ptr<float>(), dc.data_ptr<float>(),
-        reinterpret_cast<bf16*>(dq.data_ptr()),
-        (int)B, (int)H, (int)Hkv, (int)S, (int)D, (float)scale);
+    #define LAUNCH_BWD(DD)                                                    \
+        hipLaunchKernelGGL(attn_bwd_dq_kernel<DD>, dim3(grid), dim3(THREADS), \
+            lds_dq, stream,                                                   \
+            reinterpret_cast<const bf16*>(qc.data_ptr()),                     \
+            reinterpret_cast<const bf16*>(kc.data_ptr()),                     \
+            reinterpret_cast<const bf16*>(vc.data_ptr()),                     \
+            reinterpret_cast<const bf16*>(doc.data_ptr()),                    \
+            lsec.data_ptr<float>(), dc.data_ptr<float>(),                     \
+            reinterpret_cast<bf16*>(dq.data_ptr()),                           \
+            (int)B, (int)H, (int)Hkv, (int)S, (float)scale);                  \
+        hipLaunchKernelGGL(attn_bwd_dkv_kernel<DD>, dim3(grid),               \
+            dim3(THREADS), lds_dkv, stream,                                   \
+            reinterpret_cast<const bf16*>(qc.data_ptr()),                     \
+            reinterpret_cast<const bf16*>(kc.data_ptr()),                     \
+            reinterpret_cast<const bf16*>(vc.data_ptr()),                     \
+            reinterpret_cast<const bf16*>(doc.data_ptr()),                    \
+            lsec.data_ptr<float>(), dc.data_ptr<float>(),                     \
+            reinterpret_cast<bf16*>(dk.data_ptr()),                           \
+            reinterpret_cast<bf16*>(dv.data_ptr()),                           \
+            (int)B, (int)H, (int)Hkv, (int)S, (float)scale)
+    switch ((int)D) {
+        case 64: LAUNCH_BWD(64); break;
+        case 80: LAUNCH_BWD(80); break;
+        case 96: LAUNCH_BWD(96); break;
+        case 128: LAUNCH_BWD(128); break;
+        default: TORCH_CHECK(false, "head dim must be 64/80/96/128, got ", D);
+    }
+    #undef LAUNCH_BWD
     HIP_CHECK_LAST();
 
-    hipLaunchKernelGGL(attn_bwd_dkv_kernel, dim3(grid), dim3(THREADS), lds_dkv,
-        stream,
-        reinterpret_cast<const bf16*>(qc.data_ptr()),
-        reinterpret_cast<const bf16*>(kc.data_ptr()),
-        reinterpret_cast<const bf16*>(vc.data_ptr()),
-        reinterpret_cast<const bf16*>(doc.data_ptr()),
-        lsec.data_ptr<float>(), dc.data_ptr<float>(),
-        reinterpret_cast<bf16*>(dk.data_ptr()),
-        reinterpret_cast<bf16*>(dv.data_ptr()),
-        (int)B, (int)H, (int)Hkv, (int)S, (int)D, (float)scale);
-    HIP_CHECK_LAST();
     return {dq, dk, dv};
 }
