@@ -39,6 +39,10 @@ def parse_args():
     p.add_argument("--mbs", type=int, default=4, help="microbatch size")
     p.add_argument("--tp", type=int, default=1)
     p.add_argument("--pp", type=int, default=1)
+    p.add_argument("--plan-search", action="store_true",
+                   help="pick (dp, tp, pp, mbs) with the planner from "
+                        "--profile-dir instead of the flags")
+    p.add_argument("--profile-dir", default="profiles/mi355x")
     return p.parse_args()
 
 
@@ -52,8 +56,26 @@ def main() -> None:
     assert dp * args.tp * args.pp == n_gpus, "tp*pp must divide --gpus"
     gbs = args.per_gpu_batch * n_gpus // (args.tp * args.pp)
     mbs = min(args.mbs, max(gbs // dp, 1))
+    tp, pp = args.tp, args.pp
 
-    ctx = init_parallel(dp=dp, tp=args.tp, pp=args.pp)
+    est_ms = None
+    if args.plan_search and os.path.isdir(args.profile_dir):
+        from metis_amd.config import ModelConfig
+        from metis_amd.cli.plan_search import best_plan
+
+        mc = ModelConfig(
+            model_name=spec.name,
+            num_layers=spec.profile_num_layers,
+            hidden_size=spec.hidden_size,
+            sequence_length=spec.seq_length,
+            vocab_size=spec.vocab_size,
+        )
+        found = best_plan(args.profile_dir, mc, n_gpus, gbs,
+                          comm_bench_path="profiles/comm_bench.json")
+        if found:
+            dp, tp, pp, mbs, est_ms = found
+
+    ctx = init_parallel(dp=dp, tp=tp, pp=pp)
     runner = PlanRunner(spec, ctx, mbs=mbs, gbs=gbs)
 
     ms = runner.timed_steps(args.steps, args.warmup)
@@ -84,9 +106,13 @@ def main() -> None:
                 "model": args.model,
                 "global_batch": gbs,
                 "seq_len": spec.seq_length,
-                "parallelism": f"dp{dp}_tp{args.tp}_pp{args.pp}",
+                "parallelism": f"dp{dp}_tp{tp}_pp{pp}",
                 "microbatch": mbs,
                 "tokens_per_s": tokens_per_step / (ms / 1000.0),
+                "planner_estimate_ms": est_ms,
+                "cost_model_error_pct": (
+                    abs(est_ms - ms) / ms * 100.0 if est_ms else None
+                ),
             },
         }
         print(json.dumps(result))

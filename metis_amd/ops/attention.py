@@ -27,7 +27,7 @@ def _supported(q: torch.Tensor) -> bool:
         and q.dtype == torch.bfloat16
         and d % 16 == 0
         and d <= 128
-        and s % 64 == 0
+        and s % 128 == 0
     )
 
 

@@ -1,0 +1,63 @@
+"""CPU tests for the Llama family and its op fallbacks."""
+
+import torch
+
+from metis_amd.models.llama import LLAMA_SPECS, LlamaModel, LlamaModelSpec
+from metis_amd.ops.norms import apply_rope, rms_norm, swiglu
+
+
+def test_llama_forward_backward():
+    torch.manual_seed(0)
+    spec = LLAMA_SPECS["llama-tiny"]
+    model = LlamaModel(spec, dtype=torch.float32)
+    tokens = torch.randint(0, spec.vocab_size, (2, spec.seq_length))
+    loss = model(tokens, labels=torch.roll(tokens, -1, 1))
+    assert torch.isfinite(loss)
+    loss.backward()
+    assert all(p.grad is not None for p in model.parameters())
+
+
+def test_llama_stage_slicing():
+    spec = LLAMA_SPECS["llama-tiny"]
+    total = spec.profile_num_layers
+    s0 = LlamaModel(spec, dtype=torch.float32, layer_range=(0, 2))
+    s1 = LlamaModel(spec, dtype=torch.float32, layer_range=(2, total))
+    tokens = torch.randint(0, spec.vocab_size, (1, spec.seq_length))
+    hidden = s0(tokens)
+    loss = s1(hidden, labels=torch.roll(tokens, -1, 1))
+    assert torch.isfinite(loss)
+    assert len(s0.blocks) + len(s1.blocks) == spec.num_layers
+
+
+def test_llama8b_param_count():
+    spec = LLAMA_SPECS["llama3-8b"]
+    n = spec.num_parameters()
+    assert 7.5e9 < n < 8.6e9, n
+
+
+def test_rms_norm_cpu_matches_manual():
+    torch.manual_seed(1)
+    x = torch.randn(4, 64)
+    w = torch.rand(64) + 0.5
+    y = rms_norm(x, w, eps=1e-5)
+    ref = x / (x.pow(2).mean(-1, keepdim=True) + 1e-5).sqrt() * w
+    assert torch.allclose(y, ref, atol=1e-5)
+
+
+def test_rope_cpu_rotation_properties():
+    torch.manual_seed(2)
+    x = torch.randn(1, 2, 16, 32)
+    y = apply_rope(x, base=10000.0)
+    # norms preserved per rotation pair
+    d = 16
+    nx = (x[..., :d] ** 2 + x[..., d:] ** 2)
+    ny = (y[..., :d] ** 2 + y[..., d:] ** 2)
+    assert torch.allclose(nx, ny, atol=1e-4)
+    # position 0 is identity
+    assert torch.allclose(x[:, :, 0], y[:, :, 0], atol=1e-6)
+
+
+def test_swiglu_cpu():
+    a = torch.randn(32)
+    b = torch.randn(32)
+    assert torch.allclose(swiglu(a, b), torch.nn.functional.silu(a) * b)
