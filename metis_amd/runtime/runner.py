@@ -6,9 +6,9 @@ init) and reports measured iteration time, which feeds
 metis_amd.planner.validate (the cost-model-error metric).
 
 Schedule: GPipe (fill-drain), matching the cost model's
-(B-1)*max_stage + sum_stages assumption. DP gradient sync is one flat
-fp32 all-reduce after the backward passes; the fused AdamW kernel then
-consumes the flat buffer directly.
+(B-1)*max_stage + sum_stages assumption. DP gradient sync is bucketed
+and overlapped with the final backward (runtime.grad_sync); the fused
+AdamW kernel then consumes the flat fp32 buffer directly.
 """
 
 from __future__ import annotations
@@ -23,6 +23,7 @@ from metis_amd.models.gpt import GPTModel, GPTModelSpec
 from metis_amd.ops import FusedAdamW
 from metis_amd.planner.volume import uniform_layer_split
 from metis_amd.runtime.comm import ParallelContext
+from metis_amd.runtime.grad_sync import GradBucketSync
 
 
 class PlanRunner:
@@ -63,6 +64,10 @@ class PlanRunner:
             self.model.to(ctx.device)
         self.optimizer = FusedAdamW(self.model.parameters(), lr=lr)
         self.dtype = dtype
+        # bucketed, overlapped DP gradient all-reduce
+        self.grad_sync = None
+        if ctx.dp > 1 and ctx.dp_group is not None:
+            self.grad_sync = GradBucketSync(self.optimizer, ctx.dp_group, ctx.dp)
 
     # --- data -------------------------------------------------------------
     def synthetic_batch(self) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -78,7 +83,9 @@ class PlanRunner:
     def _step_no_pipeline(self) -> float:
         losses = []
         self.optimizer.zero_grad()
-        for _ in range(self.num_microbatches):
+        for mb in range(self.num_microbatches):
+            if self.grad_sync is not None and mb == self.num_microbatches - 1:
+                self.grad_sync.arm()
             tokens, labels = self.synthetic_batch()
             loss = self.model(tokens, labels=labels)
             (loss / self.num_microbatches).backward()
@@ -134,6 +141,8 @@ class PlanRunner:
 
         # backward drain (reverse order)
         for i in reversed(range(self.num_microbatches)):
+            if self.grad_sync is not None and i == 0:
+                self.grad_sync.arm()
             if ctx.is_last_stage:
                 (outputs[i] / self.num_microbatches).backward()
             else:
@@ -149,11 +158,12 @@ class PlanRunner:
 
     # --- gradient sync + optimizer ----------------------------------------
     def _sync_and_step(self) -> None:
-        ctx = self.ctx
-        grads = self.optimizer.gather_grads()
-        if ctx.dp > 1 and ctx.dp_group is not None:
-            dist.all_reduce(grads, group=ctx.dp_group)
-            grads.div_(ctx.dp)
+        if self.grad_sync is not None:
+            # hooks copied + all-reduced the final-microbatch grads,
+            # overlapped with backward; wait and average
+            self.grad_sync.finish()
+        else:
+            self.optimizer.gather_grads()
         self.optimizer.step(pre_gathered=True)
 
     def train_step(self) -> float:
