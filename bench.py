@@ -42,7 +42,8 @@ def parse_args():
     p.add_argument("--plan-search", action="store_true",
                    help="pick (dp, tp, pp, mbs) with the planner from "
                         "--profile-dir instead of the flags")
-    p.add_argument("--profile-dir", default="profiles/mi355x")
+    p.add_argument("--profile-dir", default=None,
+                   help="defaults to profiles/mi355x/<model>")
     return p.parse_args()
 
 
@@ -58,10 +59,11 @@ def main() -> None:
     mbs = min(args.mbs, max(gbs // dp, 1))
     tp, pp = args.tp, args.pp
 
+    profile_dir = args.profile_dir or f"profiles/mi355x/{args.model}"
     est_ms = None
-    if args.plan_search and os.path.isdir(args.profile_dir):
+    if os.path.isdir(profile_dir):
         from metis_amd.config import ModelConfig
-        from metis_amd.cli.plan_search import best_plan
+        from metis_amd.cli.plan_search import best_plan, estimate_plan
 
         mc = ModelConfig(
             model_name=spec.name,
@@ -70,10 +72,16 @@ def main() -> None:
             sequence_length=spec.seq_length,
             vocab_size=spec.vocab_size,
         )
-        found = best_plan(args.profile_dir, mc, n_gpus, gbs,
-                          comm_bench_path="profiles/comm_bench.json")
-        if found:
-            dp, tp, pp, mbs, est_ms = found
+        if args.plan_search:
+            found = best_plan(profile_dir, mc, n_gpus, gbs,
+                              comm_bench_path="profiles/comm_bench.json")
+            if found:
+                dp, tp, pp, mbs, est_ms = found
+        if est_ms is None:
+            # report the cost-model estimate for the plan we are running
+            est_ms = estimate_plan(profile_dir, mc, n_gpus, gbs,
+                                   dp=dp, tp=tp, pp=pp, mbs=mbs,
+                                   comm_bench_path="profiles/comm_bench.json")
 
     ctx = init_parallel(dp=dp, tp=tp, pp=pp)
     runner = PlanRunner(spec, ctx, mbs=mbs, gbs=gbs)

@@ -73,3 +73,43 @@ def best_plan(
         return None
     plan, cost = min(feasible, key=lambda r: r[1])
     return plan.dp, plan.tp, plan.pp, plan.mbs, cost
+
+
+def estimate_plan(
+    profile_dir: str,
+    model_config: ModelConfig,
+    n_gpus: int,
+    gbs: int,
+    *,
+    dp: int,
+    tp: int,
+    pp: int,
+    mbs: int,
+    device_type: str = "MI355X",
+    comm_bench_path: Optional[str] = None,
+) -> Optional[float]:
+    """Cost-model estimate (ms) for one specific plan; None if unprofiled."""
+    from metis_amd.planner.cost import HomoCostEstimator
+    from metis_amd.planner.plans import UniformPlan
+    from metis_amd.planner.volume import GPTVolume
+
+    intra, alpha = 130.0, 20.0
+    if comm_bench_path and os.path.exists(comm_bench_path):
+        with open(comm_bench_path) as fh:
+            sugg = json.load(fh).get("clusterfile_suggestion", {})
+        intra = sugg.get("intra_bandwidth", intra)
+        alpha = sugg.get("alpha_us", alpha)
+    try:
+        cluster = single_node_cluster(n_gpus, device_type, intra_bandwidth=intra)
+        store = ProfileStore.load_dir(profile_dir, optimizer_scale=1.0)
+        volume = GPTVolume(model_config, store.model.parameters_per_layer_bytes)
+        est = HomoCostEstimator(
+            store, model_config, volume, cluster,
+            PlannerArgs(gbs=gbs, max_profiled_tp_degree=max(tp, 8),
+                        max_profiled_batch_size=max(mbs, 16),
+                        comm_model="alpha_beta", alpha_us=alpha),
+        )
+        cost, _, _ = est.get_cost(UniformPlan(dp, pp, tp, mbs, gbs), device_type)
+        return cost
+    except (KeyError, FileNotFoundError):
+        return None

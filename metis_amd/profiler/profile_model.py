@@ -242,7 +242,8 @@ def main() -> None:
     p.add_argument("--model", default="gpt2-small", choices=sorted(MODEL_SPECS))
     p.add_argument("--bs", default="1,2,4,8", help="comma-separated batch sizes")
     p.add_argument("--device-type", default="MI355X")
-    p.add_argument("--out", default="profiles/mi355x")
+    p.add_argument("--out", default=None,
+                   help="defaults to profiles/mi355x/<model>")
     p.add_argument("--iters", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--seq-length", type=int, default=None)
@@ -257,10 +258,11 @@ def main() -> None:
         tp_group = ctx.tp_group
 
     spec = MODEL_SPECS[args.model]
+    out_dir = args.out or f"profiles/mi355x/{args.model}"
     for bs in [int(b) for b in args.bs.split(",")]:
         path = profile_model(
             spec, bs=bs, tp=world, device_type=args.device_type,
-            out_dir=args.out, warmup=args.warmup, iters=args.iters,
+            out_dir=out_dir, warmup=args.warmup, iters=args.iters,
             tp_group=tp_group, seq_length=args.seq_length,
         )
         if path:

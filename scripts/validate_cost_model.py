@@ -28,7 +28,7 @@ from metis_amd.runtime.comm import init_parallel  # noqa: E402
 from metis_amd.runtime.runner import PlanRunner  # noqa: E402
 
 MODEL = os.environ.get("MODEL", "gpt2-small")
-PROFILE_DIR = os.environ.get("PROFILE_DIR", "profiles/mi355x")
+PROFILE_DIR = os.environ.get("PROFILE_DIR", f"profiles/mi355x/{MODEL}")
 GBS = int(os.environ.get("GBS", "8"))
 STEPS = int(os.environ.get("STEPS", "8"))
 

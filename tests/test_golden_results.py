@@ -29,7 +29,7 @@ def test_homo_mi355x_golden():
         "--vocab_size", "51200",
         "--hostfile_path", "tests/data/mi355x_single_node/hostfile",
         "--clusterfile_path", "tests/data/mi355x_single_node/clusterfile.json",
-        "--profile_data_path", "profiles/mi355x",
+        "--profile_data_path", "profiles/mi355x/gpt2-small",
         "--max_profiled_tp_degree", "1", "--max_profiled_batch_size", "8",
         "--comm_model", "alpha_beta",
     ])
