@@ -21,7 +21,10 @@ from typing import Dict, List, Optional
 import torch
 import torch.distributed as dist
 
-from metis_amd.models.gpt import GPTModel, GPTModelSpec, MODEL_SPECS
+from metis_amd.models.gpt import GPTModel, GPTModelSpec, MODEL_SPECS as _GPT_SPECS
+from metis_amd.models.llama import LlamaModel, LlamaModelSpec, LLAMA_SPECS
+
+MODEL_SPECS = {**_GPT_SPECS, **LLAMA_SPECS}
 from metis_amd.ops import FusedAdamW
 from metis_amd.profiles import ProfileStore
 
@@ -71,32 +74,39 @@ class _LayerTimer:
 
 
 class _EmbeddingLayer(torch.nn.Module):
-    """Wraps wte+wpe as profile layer 0."""
+    """Wraps the embedding(s) as profile layer 0 (GPT: wte+wpe; Llama: wte)."""
 
-    def __init__(self, model: GPTModel):
+    def __init__(self, model):
         super().__init__()
         self.wte = model.wte
-        self.wpe = model.wpe
+        self.wpe = getattr(model, "wpe", None)
 
     def forward(self, tokens):
-        pos = torch.arange(tokens.size(1), device=tokens.device)
-        return self.wte(tokens) + self.wpe(pos)[None, :, :]
+        x = self.wte(tokens)
+        if self.wpe is not None:
+            pos = torch.arange(tokens.size(1), device=tokens.device)
+            x = x + self.wpe(pos)[None, :, :]
+        return x
 
 
 class _HeadLayer(torch.nn.Module):
-    """Wraps final LN + LM head + loss as the last profile layer."""
+    """Wraps final norm + LM head + loss as the last profile layer."""
 
-    def __init__(self, model: GPTModel):
+    def __init__(self, model):
         super().__init__()
-        self.ln_final = model.ln_final
+        self.norm = getattr(model, "ln_final", None) or model.norm_final
         self.head = model.head
         self._model = [model]  # avoid registering the full model
 
     def forward(self, x, labels):
         m = self._model[0]
-        x = self.ln_final(x)
+        x = self.norm(x)
         logits = self.head(x, m.tp_group)
-        return m._loss(logits, labels)
+        if hasattr(m, "_loss"):
+            return m._loss(logits, labels)
+        import torch.nn.functional as F
+        return F.cross_entropy(logits.float().view(-1, logits.size(-1)),
+                               labels.reshape(-1))
 
 
 def profile_model(
@@ -113,11 +123,12 @@ def profile_model(
     """Profile one (tp, bs) point; returns the JSON path (rank 0)."""
     assert torch.cuda.is_available(), "profiler needs a GPU"
     if seq_length:
-        spec = GPTModelSpec(spec.name, spec.hidden_size, spec.num_layers,
-                            spec.num_heads, spec.vocab_size, seq_length,
-                            spec.ffn_hidden_size, spec.num_kv_heads)
+        import dataclasses
+        field = "seq_length" if isinstance(spec, LlamaModelSpec) else "seq_length"
+        spec = dataclasses.replace(spec, seq_length=seq_length)
     dev = torch.device("cuda", torch.cuda.current_device())
-    model = GPTModel(spec, tp=tp, dtype=torch.bfloat16, tp_group=tp_group).to(dev)
+    model_cls = LlamaModel if isinstance(spec, LlamaModelSpec) else GPTModel
+    model = model_cls(spec, tp=tp, dtype=torch.bfloat16, tp_group=tp_group).to(dev)
     opt = FusedAdamW(model.parameters(), lr=1e-4)
 
     # profile layers: [embedding, blocks..., head]

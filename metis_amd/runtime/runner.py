@@ -20,6 +20,7 @@ import torch
 import torch.distributed as dist
 
 from metis_amd.models.gpt import GPTModel, GPTModelSpec
+from metis_amd.models.llama import LlamaModel, LlamaModelSpec
 from metis_amd.ops import FusedAdamW
 from metis_amd.planner.volume import uniform_layer_split
 from metis_amd.runtime.comm import ParallelContext
@@ -56,7 +57,8 @@ class PlanRunner:
         self.layer_partition = layer_partition
         start, end = layer_partition[ctx.pp_rank], layer_partition[ctx.pp_rank + 1]
 
-        self.model = GPTModel(
+        model_cls = LlamaModel if isinstance(spec, LlamaModelSpec) else GPTModel
+        self.model = model_cls(
             spec, tp=ctx.tp, dtype=dtype, layer_range=(start, end),
             tp_group=ctx.tp_group,
         )
