@@ -91,12 +91,14 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
 
     const int kv_end = qtile * RBLK + RBLK;
     for (int kv0 = 0; kv0 < kv_end; kv0 += CTILE) {
-        // stage K^T tile into LDS (block-wide)
+        // stage K^T tile into LDS (block-wide); lanes walk KV ROWS so the
+        // scalar transpose writes hit contiguous addresses across the wave
+        // (d-major mapping = 16-way bank conflict, see attention.hip)
         {
             constexpr int chunks = CTILE * D / 8;
             for (int c = threadIdx.x; c < chunks; c += THREADS) {
-                const int row = c / (D / 8);
-                const int d0 = (c % (D / 8)) * 8;
+                const int row = c % CTILE;
+                const int d0 = (c / CTILE) * 8;
                 bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(
                     K + kv_base + (long)(kv0 + row) * D + d0);
                 #pragma unroll
@@ -224,12 +226,13 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     // causal: only q tiles overlapping [block kv start, S)
     const int q_start = (kvtile * RBLK) / CTILE * CTILE;
     for (int q0 = q_start; q0 < S; q0 += CTILE) {
-        // stage Q^T and dO^T tiles (block-wide)
+        // stage Q^T and dO^T tiles (block-wide, kv-row-major lane walk —
+        // the d-major mapping is a 16-way bank conflict on the writes)
         {
             constexpr int chunks = CTILE * D / 8;
             for (int c = threadIdx.x; c < chunks; c += THREADS) {
-                const int row = c / (D / 8);
-                const int d0 = (c % (D / 8)) * 8;
+                const int row = c % CTILE;
+                const int d0 = (c / CTILE) * 8;
                 bf16x8 qv = *reinterpret_cast<const bf16x8*>(
                     Q + q_base + (long)(q0 + row) * D + d0);
                 bf16x8 dov = *reinterpret_cast<const bf16x8*>(
