@@ -82,7 +82,7 @@ class GradBucketSync:
                 flat[off:off + n].copy_(g.reshape(-1), non_blocking=True)
             b = self._param_bucket[idx]
             self._pending[b] -= 1
-            if self._pending[b] == 0:
+            if self._pending[b] == 0 and self.group is not None:
                 start, end, _ = self._buckets[b]
                 handle = dist.all_reduce(
                     flat[start:end], group=self.group, async_op=True
