@@ -173,6 +173,33 @@ class PlanRunner:
             return self._step_no_pipeline()
         return self._step_pipeline()
 
+    # --- checkpoint ---------------------------------------------------------
+    def save_checkpoint(self, path: str) -> None:
+        """Per-rank checkpoint (model shard + optimizer state + plan)."""
+        torch.save({
+            "model": self.model.state_dict(),
+            "optimizer": self.optimizer.state_dict(),
+            "plan": {"dp": self.ctx.dp, "tp": self.ctx.tp, "pp": self.ctx.pp,
+                     "mbs": self.mbs, "gbs": self.gbs,
+                     "layer_partition": self.layer_partition},
+            "rank": self.ctx.rank,
+        }, path)
+
+    def load_checkpoint(self, path: str) -> None:
+        state = torch.load(path, map_location="cpu", weights_only=True)
+        plan = state["plan"]
+        assert (plan["dp"], plan["tp"], plan["pp"]) == (
+            self.ctx.dp, self.ctx.tp, self.ctx.pp
+        ), "checkpoint plan does not match the running plan"
+        self.model.load_state_dict(state["model"])
+        opt_state = state["optimizer"]
+        self.optimizer.load_state_dict({
+            "step": opt_state["step"],
+            "master": opt_state["master"].to(self.optimizer.master.device),
+            "m": opt_state["m"].to(self.optimizer.m.device),
+            "v": opt_state["v"].to(self.optimizer.v.device),
+        })
+
     # --- timing -----------------------------------------------------------
     def timed_steps(self, steps: int, warmup: int) -> float:
         """Run warmup + timed steps; returns mean ms/step on this rank."""

@@ -88,3 +88,25 @@ def test_fused_adamw_matches_torch_adamw():
         mine.step()
         ref.step()
         assert torch.allclose(w.detach(), w2.detach(), atol=1e-6), step
+
+
+def test_runner_checkpoint_roundtrip(tmp_path):
+    from metis_amd.runtime.comm import ParallelContext
+    from metis_amd.runtime.runner import PlanRunner
+
+    torch.manual_seed(4)
+    spec = tiny_spec(num_layers=1)
+    ctx = ParallelContext(rank=0, world_size=1, local_rank=0, dp=1, tp=1, pp=1)
+    runner = PlanRunner(spec, ctx, mbs=2, gbs=2, dtype=torch.float32)
+    runner.train_step()
+    runner.save_checkpoint(str(tmp_path / "ckpt.pt"))
+    master_before = runner.optimizer.master.clone()
+    step_before = runner.optimizer.step_count
+
+    runner.train_step()  # diverge
+    assert not torch.equal(master_before, runner.optimizer.master)
+
+    runner.load_checkpoint(str(tmp_path / "ckpt.pt"))
+    assert torch.equal(master_before, runner.optimizer.master)
+    assert runner.optimizer.step_count == step_before
+    runner.train_step()  # resumes cleanly
