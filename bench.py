@@ -37,9 +37,9 @@ def parse_args():
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", default="gpt3-2.7b", choices=sorted(MODEL_SPECS))
-    p.add_argument("--per-gpu-batch", type=int, default=4,
+    p.add_argument("--per-gpu-batch", type=int, default=16,
                    help="sequences per GPU per step (weak scaling)")
-    p.add_argument("--mbs", type=int, default=4, help="microbatch size")
+    p.add_argument("--mbs", type=int, default=16, help="microbatch size")
     p.add_argument("--tp", type=int, default=1)
     p.add_argument("--pp", type=int, default=1)
     p.add_argument("--plan-search", action="store_true",
