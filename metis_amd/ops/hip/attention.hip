@@ -99,35 +99,56 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
         for (int jd = 0; jd < djtiles; ++jd)
             o_acc[rb][jd] = floatx4{0.f, 0.f, 0.f, 0.f};
 
-    const int kv_end = qb + QBLK;
-    for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
-        // ---- stage K (natural, padded) + V^T into LDS -------------------
-        {
-            constexpr int chunks = KVBLK * D / 8;   // 16-B chunks per tile
-            // K: lanes walk d-chunks within a row — coalesced global reads,
-            // conflict-free vector LDS writes (row stride 272 B).
-            for (int c = threadIdx.x; c < chunks; c += THREADS) {
-                const int row = c / (D / 8);
-                const int s8 = c % (D / 8);
-                bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(
-                    K + kv_base + (long)(kv0 + row) * D + s8 * 8);
-                *reinterpret_cast<bf16x8*>(Ks + (row * KSLOT + s8) * 8) = kv8;
-            }
-            // V transpose: lanes walk KV ROWS so the 8 scalar writes of a
-            // thread hit contiguous LDS addresses across the wave (the
-            // d-chunk-major mapping put all 16 lanes of a group on ONE
-            // bank: 8*VROW*2 = 1152 B stride — measured 5.5e9 conflict
-            // cycles, the kernel's dominant cost).
-            for (int c = threadIdx.x; c < chunks; c += THREADS) {
-                const int row = c % KVBLK;
-                const int s8 = c / KVBLK;
-                bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(
-                    V + kv_base + (long)(kv0 + row) * D + s8 * 8);
-                #pragma unroll
-                for (int i = 0; i < 8; ++i)
-                    Vt[(s8 * 8 + i) * VROW + row] = vv8[i];
-            }
+    // T14 async-stage split: each tile's K/V chunks are loaded to
+    // registers one tile AHEAD (HBM latency hides under the previous
+    // tile's compute); the LDS write happens at loop top from registers.
+    // 256 threads x (K: 4-row d-walk, V: kv-row walk) per the conflict
+    // analysis below.
+    constexpr int chunks = KVBLK * D / 8;           // 16-B chunks per tile
+    constexpr int per_thread = (chunks + THREADS - 1) / THREADS;
+    bf16x8 k_stage[per_thread], v_stage[per_thread];
+
+    auto issue_loads = [&](int kv0) {
+        #pragma unroll
+        for (int u = 0; u < per_thread; ++u) {
+            const int c = threadIdx.x + u * THREADS;
+            if (c >= chunks) break;
+            // K: lanes walk d-chunks within a row (coalesced reads,
+            // conflict-free vector LDS writes at row stride 272 B)
+            const int krow = c / (D / 8);
+            const int ks8 = c % (D / 8);
+            k_stage[u] = *reinterpret_cast<const bf16x8*>(
+                K + kv_base + (long)(kv0 + krow) * D + ks8 * 8);
+            // V: lanes walk KV ROWS so the scalar transpose writes hit
+            // contiguous LDS addresses (d-major mapping was a 16-way
+            // bank conflict: 5.5e9 measured conflict cycles)
+            const int vrow = c % KVBLK;
+            const int vs8 = c / KVBLK;
+            v_stage[u] = *reinterpret_cast<const bf16x8*>(
+                V + kv_base + (long)(kv0 + vrow) * D + vs8 * 8);
         }
+    };
+    auto write_stage = [&]() {
+        #pragma unroll
+        for (int u = 0; u < per_thread; ++u) {
+            const int c = threadIdx.x + u * THREADS;
+            if (c >= chunks) break;
+            const int krow = c / (D / 8);
+            const int ks8 = c % (D / 8);
+            *reinterpret_cast<bf16x8*>(Ks + (krow * KSLOT + ks8) * 8) = k_stage[u];
+            const int vrow = c % KVBLK;
+            const int vs8 = c / KVBLK;
+            #pragma unroll
+            for (int i = 0; i < 8; ++i)
+                Vt[(vs8 * 8 + i) * VROW + vrow] = v_stage[u][i];
+        }
+    };
+
+    const int kv_end = qb + QBLK;
+    issue_loads(0);
+    for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
+        write_stage();
+        if (kv0 + KVBLK < kv_end) issue_loads(kv0 + KVBLK);
         __syncthreads();
 
         // ---- per row block: S = scale * Q K^T, online softmax, P -> LDS -
@@ -139,6 +160,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
             rb_active[rb] = kv0 <= q0 + 15;
             if (!rb_active[rb]) continue;
 
+            __builtin_amdgcn_s_setprio(1);   // T5: favor the MFMA cluster
             #pragma unroll
             for (int j = 0; j < 4; ++j) {       // four 16-col subtiles
                 floatx4 s_acc = floatx4{0.f, 0.f, 0.f, 0.f};
@@ -165,6 +187,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
                     p[rb][j][r] = (kvcol > qrow) ? -1e30f : sv;
                 }
             }
+            __builtin_amdgcn_s_setprio(0);
 
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
