@@ -1,20 +1,26 @@
 // Flash attention backward for gfx950 (causal, GQA) — bf16 I/O, fp32
 // accumulation, MFMA 16x16x32, LSE-based recompute.
 //
-// Split-kernel design (deterministic, no atomics):
-//   * dQ kernel: each wave owns 16 q rows, loops kv tiles <= its rows:
-//       P  = exp(scale * Q K^T - lse)            (recompute, natural A/B)
-//       dP = dO V^T                               (both natural layouts)
-//       dS = scale * P o (dP - delta)
-//       dQ += dS @ K      (dS via per-wave LDS relayout; K^T staged in LDS)
-//   * dKV kernel: each wave owns 16 kv rows, loops q tiles >= its rows:
-//       P^T  = exp(scale * K Q^T - lse[col])
-//       dP^T = V dO^T
-//       dV  += P^T  @ dO  (P^T via LDS relayout; dO^T staged in LDS)
-//       dK  += dS^T @ Q   (dS^T via LDS relayout; Q^T staged in LDS)
-//     GQA: the kernel runs per q head into per-head dK/dV buffers; the
-//     wrapper sums each KV group (deterministic, no atomics).
-//   * delta = rowsum(dO o O) is computed by the wrapper in torch.
+// v2 structure (mirrors attention.hip's forward):
+//   * split dQ / dKV kernels (deterministic, no atomics); GQA dK/dV come
+//     back per q-head and the wrapper reduces the group;
+//   * block = 256 threads = 4 waves; each wave owns TWO 16-row blocks
+//     paired (w, 7-w) so causal work balances across waves;
+//   * opposing-side tiles are 64 wide; natural-layout A/B fragments read
+//     straight from global (the 16-32 KB tiles stay L2-hot), transposed
+//     operands (K^T for dQ; Q^T / dO^T for dK, dV) staged into padded LDS
+//     with the async T14 split: the next tile's chunks load to registers
+//     under the current tile's compute;
+//   * dS / P^T redistribute through per-wave LDS (raw bf16 bits as
+//     shorts), published by a wave-local compiler fence — DS ops of one
+//     wave complete in order;
+//   * D in {64, 80, 96, 128} template instantiations (register arrays
+//     compile-time indexed).
+//
+//   dQ kernel:  P = exp(scale*QK^T - lse); dP = dO V^T;
+//               dS = scale * P o (dP - delta); dQ += dS @ K
+//   dKV kernel: P^T = exp(scale*K Q^T - lse[col]); dP^T = V dO^T;
+//               dV += P^T @ dO; dK += dS^T @ Q
 
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
@@ -24,11 +30,11 @@
 namespace {
 
 constexpr int THREADS = 256;
-constexpr int RBLK = 64;     // rows (q or kv) per block, 16 per wave
-constexpr int CTILE = 32;    // opposing-side tile width
+constexpr int RBLK = 128;    // own-side rows per block (2 x 16 per wave)
+constexpr int CTILE = 64;    // opposing-side tile width
+constexpr int VPAD = 8;      // transposed-tile row padding (bf16)
 constexpr int DMAX = 128;
 
-// load one natural A/B fragment: 8 contiguous bf16 of `row`, chunk c
 template <int D>
 __device__ __forceinline__ bf16x8 frag8(const bf16* base, long row, int d0) {
     if (d0 < D)
@@ -39,6 +45,7 @@ __device__ __forceinline__ bf16x8 frag8(const bf16* base, long row, int d0) {
     return z;
 }
 
+// ---------------------------------------------------------------- dQ ----
 template <int D>
 __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
@@ -51,6 +58,10 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
     const int col16 = lane & 15;
     const int k8 = lane >> 4;
 
+    constexpr int dchunks = (D + 31) / 32;
+    constexpr int djtiles = D / 16;
+    constexpr int VROW = CTILE + VPAD;
+
     const int qtile = blockIdx.x % (S / RBLK);
     const int head = (blockIdx.x / (S / RBLK)) % H;
     const int batch = blockIdx.x / (S / RBLK) / H;
@@ -59,121 +70,151 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
     const long kv_base = (((long)batch * Hkv + kv_head) * S) * D;
     const long row_base = ((long)batch * H + head) * S;
 
-    const int q0 = qtile * RBLK + wave * 16;
-    constexpr int dchunks = (D + 31) / 32;
-    constexpr int djtiles = D / 16;
+    const int rbid[2] = {wave, 7 - wave};
+    const int qb = qtile * RBLK;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    short* Kt = reinterpret_cast<short*>(smem);                     // [D][32]
-    short* Sw = reinterpret_cast<short*>(smem + DMAX * CTILE * 2)
-                + wave * 16 * CTILE;                                 // [16][32]
+    short* Kt = reinterpret_cast<short*>(smem);                    // [D][VROW]
+    short* Sw = Kt + D * VROW + wave * 2 * 16 * VROW;              // [2][16][VROW]
 
-    // per-lane row state (rows k8*4 + r)
-    float lse_r[4], delta_r[4];
+    // per-row state and Q/dO fragments for both row blocks
+    float lse_r[2][4], delta_r[2][4];
+    bf16x8 q_frag[2][dchunks], do_frag[2][dchunks];
     #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        const int qrow = q0 + k8 * 4 + r;
-        lse_r[r] = LSE[row_base + qrow];
-        delta_r[r] = Delta[row_base + qrow];
+    for (int rb = 0; rb < 2; ++rb) {
+        const int q0 = qb + rbid[rb] * 16;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            lse_r[rb][r] = LSE[row_base + q0 + k8 * 4 + r];
+            delta_r[rb][r] = Delta[row_base + q0 + k8 * 4 + r];
+        }
+        #pragma unroll
+        for (int c = 0; c < dchunks; ++c) {
+            q_frag[rb][c] = frag8<D>(Q + q_base, q0 + col16, c * 32 + k8 * 8);
+            do_frag[rb][c] = frag8<D>(dO + q_base, q0 + col16, c * 32 + k8 * 8);
+        }
     }
 
-    // Q and dO fragments for this wave's rows (A layout, m = col16)
-    bf16x8 q_frag[dchunks], do_frag[dchunks];
+    floatx4 dq_acc[2][djtiles];
     #pragma unroll
-    for (int c = 0; c < dchunks; ++c) {
-        q_frag[c] = frag8<D>(Q + q_base, q0 + col16, c * 32 + k8 * 8);
-        do_frag[c] = frag8<D>(dO + q_base, q0 + col16, c * 32 + k8 * 8);
-    }
-
-    floatx4 dq_acc[djtiles];
-    #pragma unroll
-    for (int jd = 0; jd < djtiles; ++jd) dq_acc[jd] = floatx4{0.f, 0.f, 0.f, 0.f};
-
-    const int kv_end = qtile * RBLK + RBLK;
-    for (int kv0 = 0; kv0 < kv_end; kv0 += CTILE) {
-        // stage K^T tile into LDS (block-wide); lanes walk KV ROWS so the
-        // scalar transpose writes hit contiguous addresses across the wave
-        // (d-major mapping = 16-way bank conflict, see attention.hip)
-        {
-            constexpr int chunks = CTILE * D / 8;
-            for (int c = threadIdx.x; c < chunks; c += THREADS) {
-                const int row = c % CTILE;
-                const int d0 = (c / CTILE) * 8;
-                bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(
-                    K + kv_base + (long)(kv0 + row) * D + d0);
-                #pragma unroll
-                for (int i = 0; i < 8; ++i)
-                    Kt[(d0 + i) * CTILE + row] = kv8[i];
-            }
-        }
-        __syncthreads();
-        // inactive waves skip compute but keep barriers uniform
-        const bool active = kv0 <= q0 + 15;
-
-        if (active) {
-        // S and dP for two 16-col subtiles
-        floatx4 s_acc[2], dp_acc[2];
-        #pragma unroll
-        for (int j = 0; j < 2; ++j) {
-            s_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
-            dp_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
-            const long kvrow = kv0 + j * 16 + col16;
-            #pragma unroll
-            for (int c = 0; c < dchunks; ++c) {
-                const int d0 = c * 32 + k8 * 8;
-                bf16x8 k_frag = frag8<D>(K + kv_base, kvrow, d0);
-                s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    q_frag[c], k_frag, s_acc[j], 0, 0, 0);
-                bf16x8 v_frag = frag8<D>(V + kv_base, kvrow, d0);
-                dp_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    do_frag[c], v_frag, dp_acc[j], 0, 0, 0);
-            }
-        }
-
-        // dS = scale * P o (dP - delta); P = exp(scale*s - lse); mask col>row
-        #pragma unroll
-        for (int j = 0; j < 2; ++j)
-            #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                const int qrow = q0 + k8 * 4 + r;
-                const int kvcol = kv0 + j * 16 + col16;
-                float p = (kvcol > qrow)
-                              ? 0.f
-                              : __expf(s_acc[j][r] * scale - lse_r[r]);
-                float ds = scale * p * (dp_acc[j][r] - delta_r[r]);
-                Sw[(k8 * 4 + r) * CTILE + j * 16 + col16] = float_to_bf16_bits(ds);
-            }
-        }  // active
-
-        // barrier orders the scalar dS stores against the vector re-read
-        // (different pointer types: TBAA would otherwise allow hoisting)
-        __syncthreads();
-
-        if (active) {
-        // dQ += dS @ K : A = dS (LDS relayout), B = K^T from LDS
-        bf16x8 ds_frag = *reinterpret_cast<const bf16x8*>(
-            Sw + col16 * CTILE + k8 * 8);
-        #pragma unroll
-        for (int jd = 0; jd < djtiles; ++jd) {
-            bf16x8 kt_frag = *reinterpret_cast<const bf16x8*>(
-                Kt + (jd * 16 + col16) * CTILE + k8 * 8);
-            dq_acc[jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                ds_frag, kt_frag, dq_acc[jd], 0, 0, 0);
-        }
-        }  // active
-        __syncthreads();
-    }
-
-    #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        const long qrow = q0 + k8 * 4 + r;
+    for (int rb = 0; rb < 2; ++rb)
         #pragma unroll
         for (int jd = 0; jd < djtiles; ++jd)
-            dQ[q_base + qrow * D + jd * 16 + col16] =
-                __float2bfloat16(dq_acc[jd][r]);
+            dq_acc[rb][jd] = floatx4{0.f, 0.f, 0.f, 0.f};
+
+    // T14 staged K^T: chunks loaded to registers one tile ahead,
+    // kv-row-major lane walk (conflict-free transpose writes)
+    constexpr int chunks = CTILE * D / 8;
+    constexpr int per_thread = (chunks + THREADS - 1) / THREADS;
+    bf16x8 k_stage[per_thread];
+    auto issue_loads = [&](int kv0) {
+        #pragma unroll
+        for (int u = 0; u < per_thread; ++u) {
+            const int c = threadIdx.x + u * THREADS;
+            if (c >= chunks) break;
+            k_stage[u] = *reinterpret_cast<const bf16x8*>(
+                K + kv_base + (long)(kv0 + c % CTILE) * D + (c / CTILE) * 8);
+        }
+    };
+    auto write_stage = [&]() {
+        #pragma unroll
+        for (int u = 0; u < per_thread; ++u) {
+            const int c = threadIdx.x + u * THREADS;
+            if (c >= chunks) break;
+            const int row = c % CTILE;
+            const int d0 = (c / CTILE) * 8;
+            #pragma unroll
+            for (int i = 0; i < 8; ++i)
+                Kt[(d0 + i) * VROW + row] = k_stage[u][i];
+        }
+    };
+
+    const int kv_end = qb + RBLK;
+    issue_loads(0);
+    for (int kv0 = 0; kv0 < kv_end; kv0 += CTILE) {
+        write_stage();
+        if (kv0 + CTILE < kv_end) issue_loads(kv0 + CTILE);
+        __syncthreads();
+
+        bool rb_active[2];
+        #pragma unroll
+        for (int rb = 0; rb < 2; ++rb) {
+            const int q0 = qb + rbid[rb] * 16;
+            rb_active[rb] = kv0 <= q0 + 15;
+            if (!rb_active[rb]) continue;
+
+            __builtin_amdgcn_s_setprio(1);
+            floatx4 s_acc[4], dp_acc[4];
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                s_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
+                dp_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
+                const long kvrow = kv0 + j * 16 + col16;
+                #pragma unroll
+                for (int c = 0; c < dchunks; ++c) {
+                    const int d0 = c * 32 + k8 * 8;
+                    s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        q_frag[rb][c], frag8<D>(K + kv_base, kvrow, d0),
+                        s_acc[j], 0, 0, 0);
+                    dp_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        do_frag[rb][c], frag8<D>(V + kv_base, kvrow, d0),
+                        dp_acc[j], 0, 0, 0);
+                }
+            }
+            __builtin_amdgcn_s_setprio(0);
+
+            short* Srb = Sw + rb * 16 * VROW;
+            #pragma unroll
+            for (int j = 0; j < 4; ++j)
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int qrow = q0 + k8 * 4 + r;
+                    const int kvcol = kv0 + j * 16 + col16;
+                    float p = (kvcol > qrow)
+                                  ? 0.f
+                                  : __expf(s_acc[j][r] * scale - lse_r[rb][r]);
+                    float ds = scale * p * (dp_acc[j][r] - delta_r[rb][r]);
+                    Srb[(k8 * 4 + r) * VROW + j * 16 + col16] =
+                        float_to_bf16_bits(ds);
+                }
+        }
+
+        // wave-local publish of dS (DS ops of one wave are in order)
+        asm volatile("" ::: "memory");
+
+        #pragma unroll
+        for (int rb = 0; rb < 2; ++rb) {
+            if (!rb_active[rb]) continue;
+            const short* Srb = Sw + rb * 16 * VROW;
+            #pragma unroll
+            for (int ks = 0; ks < 2; ++ks) {
+                bf16x8 ds_frag = *reinterpret_cast<const bf16x8*>(
+                    Srb + col16 * VROW + ks * 32 + k8 * 8);
+                #pragma unroll
+                for (int jd = 0; jd < djtiles; ++jd) {
+                    bf16x8 kt_frag = *reinterpret_cast<const bf16x8*>(
+                        Kt + (jd * 16 + col16) * VROW + ks * 32 + k8 * 8);
+                    dq_acc[rb][jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        ds_frag, kt_frag, dq_acc[rb][jd], 0, 0, 0);
+                }
+            }
+        }
+        __syncthreads();
     }
+
+    #pragma unroll
+    for (int rb = 0; rb < 2; ++rb)
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const long qrow = qb + rbid[rb] * 16 + k8 * 4 + r;
+            #pragma unroll
+            for (int jd = 0; jd < djtiles; ++jd)
+                dQ[q_base + qrow * D + jd * 16 + col16] =
+                    __float2bfloat16(dq_acc[rb][jd][r]);
+        }
 }
 
+// --------------------------------------------------------------- dKV ----
 template <int D>
 __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
@@ -187,6 +228,10 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     const int col16 = lane & 15;
     const int k8 = lane >> 4;
 
+    constexpr int dchunks = (D + 31) / 32;
+    constexpr int djtiles = D / 16;
+    constexpr int VROW = CTILE + VPAD;
+
     const int kvtile = blockIdx.x % (S / RBLK);
     const int head = (blockIdx.x / (S / RBLK)) % H;
     const int batch = blockIdx.x / (S / RBLK) / H;
@@ -196,144 +241,151 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     const long out_base = (((long)batch * H + head) * S) * D;
     const long row_base = ((long)batch * H + head) * S;
 
-    const int kv0 = kvtile * RBLK + wave * 16;   // wave's first kv row
-    constexpr int dchunks = (D + 31) / 32;
-    constexpr int djtiles = D / 16;
+    const int rbid[2] = {wave, 7 - wave};
+    const int kb = kvtile * RBLK;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    short* Qt = reinterpret_cast<short*>(smem);                      // [D][32]
-    short* dOt = reinterpret_cast<short*>(smem + DMAX * CTILE * 2);   // [D][32]
-    short* Sw = reinterpret_cast<short*>(smem + 2 * DMAX * CTILE * 2)
-                + wave * 16 * CTILE;                                  // [16][32]
-    short* Pw = reinterpret_cast<short*>(smem + 2 * DMAX * CTILE * 2
-                + 4 * 16 * CTILE * 2) + wave * 16 * CTILE;            // [16][32]
+    short* Qt = reinterpret_cast<short*>(smem);                    // [D][VROW]
+    short* dOt = Qt + D * VROW;                                    // [D][VROW]
+    short* Sw = dOt + D * VROW + wave * 2 * 16 * VROW;             // [2][16][VROW]
+    short* Pw = dOt + D * VROW + 8 * 16 * VROW + wave * 2 * 16 * VROW;
 
-    // K and V fragments for this wave's rows (A layout, m = col16)
-    bf16x8 k_frag[dchunks], v_frag[dchunks];
+    // K and V fragments for both row blocks (A layout, m = col16)
+    bf16x8 k_frag[2][dchunks], v_frag[2][dchunks];
     #pragma unroll
-    for (int c = 0; c < dchunks; ++c) {
-        k_frag[c] = frag8<D>(K + kv_base, kv0 + col16, c * 32 + k8 * 8);
-        v_frag[c] = frag8<D>(V + kv_base, kv0 + col16, c * 32 + k8 * 8);
+    for (int rb = 0; rb < 2; ++rb) {
+        const int kv0 = kb + rbid[rb] * 16;
+        #pragma unroll
+        for (int c = 0; c < dchunks; ++c) {
+            k_frag[rb][c] = frag8<D>(K + kv_base, kv0 + col16, c * 32 + k8 * 8);
+            v_frag[rb][c] = frag8<D>(V + kv_base, kv0 + col16, c * 32 + k8 * 8);
+        }
     }
 
-    floatx4 dk_acc[djtiles], dv_acc[djtiles];
+    floatx4 dk_acc[2][djtiles], dv_acc[2][djtiles];
     #pragma unroll
-    for (int jd = 0; jd < djtiles; ++jd) {
-        dk_acc[jd] = floatx4{0.f, 0.f, 0.f, 0.f};
-        dv_acc[jd] = floatx4{0.f, 0.f, 0.f, 0.f};
-    }
+    for (int rb = 0; rb < 2; ++rb)
+        #pragma unroll
+        for (int jd = 0; jd < djtiles; ++jd) {
+            dk_acc[rb][jd] = floatx4{0.f, 0.f, 0.f, 0.f};
+            dv_acc[rb][jd] = floatx4{0.f, 0.f, 0.f, 0.f};
+        }
 
-    // causal: only q tiles overlapping [block kv start, S)
-    const int q_start = (kvtile * RBLK) / CTILE * CTILE;
+    // Q^T / dO^T staged straight from global each tile (dkv already
+    // carries 128 accumulator VGPRs for dK+dV; holding T14 stage
+    // registers on top spills — measured 105 spilled VGPRs at D=128)
+    constexpr int chunks = CTILE * D / 8;
+    auto stage_tile = [&](int q0) {
+        for (int c = threadIdx.x; c < chunks; c += THREADS) {
+            const int row = c % CTILE;
+            const int d0 = (c / CTILE) * 8;
+            bf16x8 qv = *reinterpret_cast<const bf16x8*>(
+                Q + q_base + (long)(q0 + row) * D + d0);
+            bf16x8 dov = *reinterpret_cast<const bf16x8*>(
+                dO + q_base + (long)(q0 + row) * D + d0);
+            #pragma unroll
+            for (int i = 0; i < 8; ++i) {
+                Qt[(d0 + i) * VROW + row] = qv[i];
+                dOt[(d0 + i) * VROW + row] = dov[i];
+            }
+        }
+    };
+
+    const int q_start = kb;    // causal: q tiles from the block's kv start
     for (int q0 = q_start; q0 < S; q0 += CTILE) {
-        // stage Q^T and dO^T tiles (block-wide, kv-row-major lane walk —
-        // the d-major mapping is a 16-way bank conflict on the writes)
-        {
-            constexpr int chunks = CTILE * D / 8;
-            for (int c = threadIdx.x; c < chunks; c += THREADS) {
-                const int row = c % CTILE;
-                const int d0 = (c / CTILE) * 8;
-                bf16x8 qv = *reinterpret_cast<const bf16x8*>(
-                    Q + q_base + (long)(q0 + row) * D + d0);
-                bf16x8 dov = *reinterpret_cast<const bf16x8*>(
-                    dO + q_base + (long)(q0 + row) * D + d0);
+        stage_tile(q0);
+        __syncthreads();
+
+        bool rb_active[2];
+        #pragma unroll
+        for (int rb = 0; rb < 2; ++rb) {
+            const int kv0 = kb + rbid[rb] * 16;
+            rb_active[rb] = q0 + CTILE - 1 >= kv0;
+            if (!rb_active[rb]) continue;
+
+            __builtin_amdgcn_s_setprio(1);
+            floatx4 st_acc[4], dpt_acc[4];
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                st_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
+                dpt_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
+                const long qrow = q0 + j * 16 + col16;
                 #pragma unroll
-                for (int i = 0; i < 8; ++i) {
-                    Qt[(d0 + i) * CTILE + row] = qv[i];
-                    dOt[(d0 + i) * CTILE + row] = dov[i];
+                for (int c = 0; c < dchunks; ++c) {
+                    const int d0 = c * 32 + k8 * 8;
+                    st_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        k_frag[rb][c], frag8<D>(Q + q_base, qrow, d0),
+                        st_acc[j], 0, 0, 0);
+                    dpt_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        v_frag[rb][c], frag8<D>(dO + q_base, qrow, d0),
+                        dpt_acc[j], 0, 0, 0);
+                }
+            }
+            __builtin_amdgcn_s_setprio(0);
+
+            short* Srb = Sw + rb * 16 * VROW;
+            short* Prb = Pw + rb * 16 * VROW;
+            #pragma unroll
+            for (int j = 0; j < 4; ++j)
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int kvrow = kb + rbid[rb] * 16 + k8 * 4 + r;
+                    const int qcol = q0 + j * 16 + col16;
+                    float p = (qcol < kvrow)
+                                  ? 0.f
+                                  : __expf(st_acc[j][r] * scale
+                                           - LSE[row_base + qcol]);
+                    float ds = scale * p
+                               * (dpt_acc[j][r] - Delta[row_base + qcol]);
+                    Srb[(k8 * 4 + r) * VROW + j * 16 + col16] =
+                        float_to_bf16_bits(ds);
+                    Prb[(k8 * 4 + r) * VROW + j * 16 + col16] =
+                        float_to_bf16_bits(p);
+                }
+        }
+
+        asm volatile("" ::: "memory");   // wave-local dS^T / P^T publish
+
+        #pragma unroll
+        for (int rb = 0; rb < 2; ++rb) {
+            if (!rb_active[rb]) continue;
+            const short* Srb = Sw + rb * 16 * VROW;
+            const short* Prb = Pw + rb * 16 * VROW;
+            #pragma unroll
+            for (int ks = 0; ks < 2; ++ks) {
+                bf16x8 dst_frag = *reinterpret_cast<const bf16x8*>(
+                    Srb + col16 * VROW + ks * 32 + k8 * 8);
+                bf16x8 pt_frag = *reinterpret_cast<const bf16x8*>(
+                    Prb + col16 * VROW + ks * 32 + k8 * 8);
+                #pragma unroll
+                for (int jd = 0; jd < djtiles; ++jd) {
+                    bf16x8 qt_frag = *reinterpret_cast<const bf16x8*>(
+                        Qt + (jd * 16 + col16) * VROW + ks * 32 + k8 * 8);
+                    dk_acc[rb][jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        dst_frag, qt_frag, dk_acc[rb][jd], 0, 0, 0);
+                    bf16x8 dot_frag = *reinterpret_cast<const bf16x8*>(
+                        dOt + (jd * 16 + col16) * VROW + ks * 32 + k8 * 8);
+                    dv_acc[rb][jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        pt_frag, dot_frag, dv_acc[rb][jd], 0, 0, 0);
                 }
             }
         }
         __syncthreads();
-        // entire q tile above the diagonal -> inactive (barriers uniform)
-        const bool active = q0 + CTILE - 1 >= kv0;
-
-        float pt[2][4];
-        if (active) {
-        // S^T = K Q^T and dP^T = V dO^T for two 16-col (q) subtiles
-        floatx4 st_acc[2], dpt_acc[2];
-        #pragma unroll
-        for (int j = 0; j < 2; ++j) {
-            st_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
-            dpt_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
-            const long qrow = q0 + j * 16 + col16;
-            #pragma unroll
-            for (int c = 0; c < dchunks; ++c) {
-                const int d0 = c * 32 + k8 * 8;
-                bf16x8 qf = frag8<D>(Q + q_base, qrow, d0);
-                st_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    k_frag[c], qf, st_acc[j], 0, 0, 0);
-                bf16x8 dof = frag8<D>(dO + q_base, qrow, d0);
-                dpt_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    v_frag[c], dof, dpt_acc[j], 0, 0, 0);
-            }
-        }
-
-        // P^T and dS^T (C layout: row = kv = k8*4+r, col = q)
-        #pragma unroll
-        for (int j = 0; j < 2; ++j)
-            #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                const int kvrow = kv0 + k8 * 4 + r;
-                const int qcol = q0 + j * 16 + col16;
-                float p = (qcol < kvrow)
-                              ? 0.f
-                              : __expf(st_acc[j][r] * scale
-                                       - LSE[row_base + qcol]);
-                pt[j][r] = p;
-                float ds = scale * p
-                           * (dpt_acc[j][r] - Delta[row_base + qcol]);
-                Sw[(k8 * 4 + r) * CTILE + j * 16 + col16] = float_to_bf16_bits(ds);
-                Pw[(k8 * 4 + r) * CTILE + j * 16 + col16] =
-                    float_to_bf16_bits(pt[j][r]);
-            }
-        }  // active
-
-        // barrier orders the scalar dS^T / P^T stores against the vector
-        // re-reads (TBAA) and keeps control flow uniform
-        __syncthreads();
-
-        if (active) {
-        // dK += dS^T @ Q : A = dS^T (LDS relayout), B = Q^T (LDS)
-        {
-            bf16x8 dst_frag = *reinterpret_cast<const bf16x8*>(
-                Sw + col16 * CTILE + k8 * 8);
-            #pragma unroll
-            for (int jd = 0; jd < djtiles; ++jd) {
-                bf16x8 qt_frag = *reinterpret_cast<const bf16x8*>(
-                    Qt + (jd * 16 + col16) * CTILE + k8 * 8);
-                dk_acc[jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    dst_frag, qt_frag, dk_acc[jd], 0, 0, 0);
-            }
-        }
-
-        // dV += P^T @ dO : A = P^T (LDS relayout), B = dO^T (LDS)
-        {
-            bf16x8 pt_frag = *reinterpret_cast<const bf16x8*>(
-                Pw + col16 * CTILE + k8 * 8);
-            #pragma unroll
-            for (int jd = 0; jd < djtiles; ++jd) {
-                bf16x8 dot_frag = *reinterpret_cast<const bf16x8*>(
-                    dOt + (jd * 16 + col16) * CTILE + k8 * 8);
-                dv_acc[jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    pt_frag, dot_frag, dv_acc[jd], 0, 0, 0);
-            }
-        }
-        }  // active
-        __syncthreads();
     }
 
     #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        const long kvrow = kv0 + k8 * 4 + r;
+    for (int rb = 0; rb < 2; ++rb)
         #pragma unroll
-        for (int jd = 0; jd < djtiles; ++jd) {
-            dK[out_base + kvrow * D + jd * 16 + col16] =
-                __float2bfloat16(dk_acc[jd][r]);
-            dV[out_base + kvrow * D + jd * 16 + col16] =
-                __float2bfloat16(dv_acc[jd][r]);
+        for (int r = 0; r < 4; ++r) {
+            const long kvrow = kb + rbid[rb] * 16 + k8 * 4 + r;
+            #pragma unroll
+            for (int jd = 0; jd < djtiles; ++jd) {
+                dK[out_base + kvrow * D + jd * 16 + col16] =
+                    __float2bfloat16(dk_acc[rb][jd][r]);
+                dV[out_base + kvrow * D + jd * 16 + col16] =
+                    __float2bfloat16(dv_acc[rb][jd][r]);
+            }
         }
-    }
 }
 
 }  // namespace
@@ -345,41 +397,43 @@ std::vector<torch::Tensor> attn_bwd(
     TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
     const long B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
     const long Hkv = k.size(1);
-    TORCH_CHECK(S % RBLK == 0 && D % 16 == 0 && D <= DMAX);
+    TORCH_CHECK(S % RBLK == 0, "sequence length must be a multiple of 128");
     auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
     auto doc = d_o.contiguous();
     auto lsec = lse.contiguous(), dc = delta.contiguous();
 
     auto dq = torch::empty_like(qc);
-    // per q-head dK/dV; the wrapper reduces the GQA group
     auto dk = torch::empty({B, H, S, D}, q.options());
     auto dv = torch::empty({B, H, S, D}, q.options());
 
     auto stream = c10::hip::getCurrentHIPStream().stream();
     const int grid = (int)(B * H * (S / RBLK));
-    const int lds_dq = DMAX * CTILE * 2 + 4 * 16 * CTILE * 2;
-    const int lds_dkv = 2 * DMAX * CTILE * 2 + 2 * 4 * 16 * CTILE * 2;
 
     #define LAUNCH_BWD(DD)                                                    \
-        hipLaunchKernelGGL(attn_bwd_dq_kernel<DD>, dim3(grid), dim3(THREADS), \
-            lds_dq, stream,                                                   \
-            reinterpret_cast<const bf16*>(qc.data_ptr()),                     \
-            reinterpret_cast<const bf16*>(kc.data_ptr()),                     \
-            reinterpret_cast<const bf16*>(vc.data_ptr()),                     \
-            reinterpret_cast<const bf16*>(doc.data_ptr()),                    \
-            lsec.data_ptr<float>(), dc.data_ptr<float>(),                     \
-            reinterpret_cast<bf16*>(dq.data_ptr()),                           \
-            (int)B, (int)H, (int)Hkv, (int)S, (float)scale);                  \
-        hipLaunchKernelGGL(attn_bwd_dkv_kernel<DD>, dim3(grid),               \
-            dim3(THREADS), lds_dkv, stream,                                   \
-            reinterpret_cast<const bf16*>(qc.data_ptr()),                     \
-            reinterpret_cast<const bf16*>(kc.data_ptr()),                     \
-            reinterpret_cast<const bf16*>(vc.data_ptr()),                     \
-            reinterpret_cast<const bf16*>(doc.data_ptr()),                    \
-            lsec.data_ptr<float>(), dc.data_ptr<float>(),                     \
-            reinterpret_cast<bf16*>(dk.data_ptr()),                           \
-            reinterpret_cast<bf16*>(dv.data_ptr()),                           \
-            (int)B, (int)H, (int)Hkv, (int)S, (float)scale)
+        do {                                                                  \
+            const int vrow = CTILE + VPAD;                                    \
+            const int lds_dq = (DD * vrow + 4 * 2 * 16 * vrow) * 2;           \
+            const int lds_dkv = (2 * DD * vrow + 2 * 8 * 16 * vrow) * 2;      \
+            hipLaunchKernelGGL(attn_bwd_dq_kernel<DD>, dim3(grid),            \
+                dim3(THREADS), lds_dq, stream,                                \
+                reinterpret_cast<const bf16*>(qc.data_ptr()),                 \
+                reinterpret_cast<const bf16*>(kc.data_ptr()),                 \
+                reinterpret_cast<const bf16*>(vc.data_ptr()),                 \
+                reinterpret_cast<const bf16*>(doc.data_ptr()),                \
+                lsec.data_ptr<float>(), dc.data_ptr<float>(),                 \
+                reinterpret_cast<bf16*>(dq.data_ptr()),                       \
+                (int)B, (int)H, (int)Hkv, (int)S, (float)scale);              \
+            hipLaunchKernelGGL(attn_bwd_dkv_kernel<DD>, dim3(grid),           \
+                dim3(THREADS), lds_dkv, stream,                               \
+                reinterpret_cast<const bf16*>(qc.data_ptr()),                 \
+                reinterpret_cast<const bf16*>(kc.data_ptr()),                 \
+                reinterpret_cast<const bf16*>(vc.data_ptr()),                 \
+                reinterpret_cast<const bf16*>(doc.data_ptr()),                \
+                lsec.data_ptr<float>(), dc.data_ptr<float>(),                 \
+                reinterpret_cast<bf16*>(dk.data_ptr()),                       \
+                reinterpret_cast<bf16*>(dv.data_ptr()),                       \
+                (int)B, (int)H, (int)Hkv, (int)S, (float)scale);              \
+        } while (0)
     switch ((int)D) {
         case 64: LAUNCH_BWD(64); break;
         case 80: LAUNCH_BWD(80); break;
@@ -389,6 +443,5 @@ std::vector<torch::Tensor> attn_bwd(
     }
     #undef LAUNCH_BWD
     HIP_CHECK_LAST();
-
     return {dq, dk, dv};
 }
