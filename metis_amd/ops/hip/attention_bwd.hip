@@ -73,8 +73,11 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
     const int rbid[2] = {wave, 7 - wave};
     const int qb = qtile * RBLK;
 
+    constexpr int KSLOT = D / 8 + 1;    // natural-tile slots per row (padded)
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    short* Kt = reinterpret_cast<short*>(smem);                    // [D][VROW]
+    short* Ks = reinterpret_cast<short*>(smem);                    // [CTILE][KSLOT*8]
+    short* Vs = Ks + CTILE * KSLOT * 8;                            // [CTILE][KSLOT*8]
+    short* Kt = Vs + CTILE * KSLOT * 8;                            // [D][VROW]
     short* Sw = Kt + D * VROW + wave * 2 * 16 * VROW;              // [2][16][VROW]
 
     // per-row state and Q/dO fragments for both row blocks
@@ -102,8 +105,9 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
         for (int jd = 0; jd < djtiles; ++jd)
             dq_acc[rb][jd] = floatx4{0.f, 0.f, 0.f, 0.f};
 
-    // T14 staged K^T: chunks loaded to registers one tile ahead,
-    // kv-row-major lane walk (conflict-free transpose writes)
+    // T14 staged K (reused for the natural tile AND the transpose) plus
+    // a direct-staged natural V tile; kv-row-major lane walk keeps the
+    // transpose writes conflict-free
     constexpr int chunks = CTILE * D / 8;
     constexpr int per_thread = (chunks + THREADS - 1) / THREADS;
     bf16x8 k_stage[per_thread];
@@ -116,23 +120,27 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
                 K + kv_base + (long)(kv0 + c % CTILE) * D + (c / CTILE) * 8);
         }
     };
-    auto write_stage = [&]() {
+    auto write_stage = [&](int kv0) {
         #pragma unroll
         for (int u = 0; u < per_thread; ++u) {
             const int c = threadIdx.x + u * THREADS;
             if (c >= chunks) break;
             const int row = c % CTILE;
             const int d0 = (c / CTILE) * 8;
+            *reinterpret_cast<bf16x8*>(Ks + row * KSLOT * 8 + d0) = k_stage[u];
             #pragma unroll
             for (int i = 0; i < 8; ++i)
                 Kt[(d0 + i) * VROW + row] = k_stage[u][i];
+            bf16x8 vv = *reinterpret_cast<const bf16x8*>(
+                V + kv_base + (long)(kv0 + row) * D + d0);
+            *reinterpret_cast<bf16x8*>(Vs + row * KSLOT * 8 + d0) = vv;
         }
     };
 
     const int kv_end = qb + RBLK;
     issue_loads(0);
     for (int kv0 = 0; kv0 < kv_end; kv0 += CTILE) {
-        write_stage();
+        write_stage(kv0);
         if (kv0 + CTILE < kv_end) issue_loads(kv0 + CTILE);
         __syncthreads();
 
@@ -149,16 +157,24 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
             for (int j = 0; j < 4; ++j) {
                 s_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
                 dp_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
-                const long kvrow = kv0 + j * 16 + col16;
+                const int kvrow = j * 16 + col16;
                 #pragma unroll
                 for (int c = 0; c < dchunks; ++c) {
                     const int d0 = c * 32 + k8 * 8;
+                    bf16x8 kf, vf;
+                    if (d0 < D) {
+                        kf = *reinterpret_cast<const bf16x8*>(
+                            Ks + kvrow * KSLOT * 8 + d0);
+                        vf = *reinterpret_cast<const bf16x8*>(
+                            Vs + kvrow * KSLOT * 8 + d0);
+                    } else {
+                        #pragma unroll
+                        for (int i = 0; i < 8; ++i) { kf[i] = 0; vf[i] = 0; }
+                    }
                     s_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        q_frag[rb][c], frag8<D>(K + kv_base, kvrow, d0),
-                        s_acc[j], 0, 0, 0);
+                        q_frag[rb][c], kf, s_acc[j], 0, 0, 0);
                     dp_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        do_frag[rb][c], frag8<D>(V + kv_base, kvrow, d0),
-                        dp_acc[j], 0, 0, 0);
+                        do_frag[rb][c], vf, dp_acc[j], 0, 0, 0);
                 }
             }
             __builtin_amdgcn_s_setprio(0);
@@ -215,6 +231,9 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
 }
 
 // --------------------------------------------------------------- dKV ----
+// One 16-row kv block per wave (64 per block): dkv's causal skip is at
+// most one q tile per wave, so row-block pairing buys nothing and the
+// freed registers/LDS go to staged natural Q/dO tiles instead.
 template <int D>
 __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
@@ -231,105 +250,119 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     constexpr int dchunks = (D + 31) / 32;
     constexpr int djtiles = D / 16;
     constexpr int VROW = CTILE + VPAD;
+    constexpr int KSLOT = D / 8 + 1;
+    constexpr int KVB = 64;            // kv rows per block (16 per wave)
 
-    const int kvtile = blockIdx.x % (S / RBLK);
-    const int head = (blockIdx.x / (S / RBLK)) % H;
-    const int batch = blockIdx.x / (S / RBLK) / H;
+    const int kvtile = blockIdx.x % (S / KVB);
+    const int head = (blockIdx.x / (S / KVB)) % H;
+    const int batch = blockIdx.x / (S / KVB) / H;
     const int kv_head = head / (H / Hkv);
     const long q_base = (((long)batch * H + head) * S) * D;
     const long kv_base = (((long)batch * Hkv + kv_head) * S) * D;
     const long out_base = (((long)batch * H + head) * S) * D;
     const long row_base = ((long)batch * H + head) * S;
 
-    const int rbid[2] = {wave, 7 - wave};
-    const int kb = kvtile * RBLK;
+    const int kv0 = kvtile * KVB + wave * 16;   // this wave's kv rows
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    short* Qt = reinterpret_cast<short*>(smem);                    // [D][VROW]
+    short* Qs = reinterpret_cast<short*>(smem);                    // [CTILE][KSLOT*8]
+    short* dOs = Qs + CTILE * KSLOT * 8;                           // [CTILE][KSLOT*8]
+    short* Qt = dOs + CTILE * KSLOT * 8;                           // [D][VROW]
     short* dOt = Qt + D * VROW;                                    // [D][VROW]
-    short* Sw = dOt + D * VROW + wave * 2 * 16 * VROW;             // [2][16][VROW]
-    short* Pw = dOt + D * VROW + 8 * 16 * VROW + wave * 2 * 16 * VROW;
+    short* Sw = dOt + D * VROW + wave * 16 * VROW;                 // [16][VROW]
+    short* Pw = dOt + D * VROW + 4 * 16 * VROW + wave * 16 * VROW; // [16][VROW]
 
-    // K and V fragments for both row blocks (A layout, m = col16)
-    bf16x8 k_frag[2][dchunks], v_frag[2][dchunks];
+    // K and V fragments for this wave's rows (A layout, m = col16)
+    bf16x8 k_frag[dchunks], v_frag[dchunks];
     #pragma unroll
-    for (int rb = 0; rb < 2; ++rb) {
-        const int kv0 = kb + rbid[rb] * 16;
-        #pragma unroll
-        for (int c = 0; c < dchunks; ++c) {
-            k_frag[rb][c] = frag8<D>(K + kv_base, kv0 + col16, c * 32 + k8 * 8);
-            v_frag[rb][c] = frag8<D>(V + kv_base, kv0 + col16, c * 32 + k8 * 8);
-        }
+    for (int c = 0; c < dchunks; ++c) {
+        k_frag[c] = frag8<D>(K + kv_base, kv0 + col16, c * 32 + k8 * 8);
+        v_frag[c] = frag8<D>(V + kv_base, kv0 + col16, c * 32 + k8 * 8);
     }
 
-    floatx4 dk_acc[2][djtiles], dv_acc[2][djtiles];
+    floatx4 dk_acc[djtiles], dv_acc[djtiles];
     #pragma unroll
-    for (int rb = 0; rb < 2; ++rb)
-        #pragma unroll
-        for (int jd = 0; jd < djtiles; ++jd) {
-            dk_acc[rb][jd] = floatx4{0.f, 0.f, 0.f, 0.f};
-            dv_acc[rb][jd] = floatx4{0.f, 0.f, 0.f, 0.f};
-        }
+    for (int jd = 0; jd < djtiles; ++jd) {
+        dk_acc[jd] = floatx4{0.f, 0.f, 0.f, 0.f};
+        dv_acc[jd] = floatx4{0.f, 0.f, 0.f, 0.f};
+    }
 
-    // Q^T / dO^T staged straight from global each tile (dkv already
-    // carries 128 accumulator VGPRs for dK+dV; holding T14 stage
-    // registers on top spills — measured 105 spilled VGPRs at D=128)
+    // T14 staged Q (natural + transpose from one register set) plus a
+    // direct-staged dO pair; q-row-major lane walk
     constexpr int chunks = CTILE * D / 8;
-    auto stage_tile = [&](int q0) {
-        for (int c = threadIdx.x; c < chunks; c += THREADS) {
+    constexpr int per_thread = (chunks + THREADS - 1) / THREADS;
+    bf16x8 q_stage[per_thread];
+    auto issue_loads = [&](int q0) {
+        #pragma unroll
+        for (int u = 0; u < per_thread; ++u) {
+            const int c = threadIdx.x + u * THREADS;
+            if (c >= chunks) break;
+            q_stage[u] = *reinterpret_cast<const bf16x8*>(
+                Q + q_base + (long)(q0 + c % CTILE) * D + (c / CTILE) * 8);
+        }
+    };
+    auto write_stage = [&](int q0) {
+        #pragma unroll
+        for (int u = 0; u < per_thread; ++u) {
+            const int c = threadIdx.x + u * THREADS;
+            if (c >= chunks) break;
             const int row = c % CTILE;
             const int d0 = (c / CTILE) * 8;
-            bf16x8 qv = *reinterpret_cast<const bf16x8*>(
-                Q + q_base + (long)(q0 + row) * D + d0);
+            *reinterpret_cast<bf16x8*>(Qs + row * KSLOT * 8 + d0) = q_stage[u];
+            #pragma unroll
+            for (int i = 0; i < 8; ++i)
+                Qt[(d0 + i) * VROW + row] = q_stage[u][i];
             bf16x8 dov = *reinterpret_cast<const bf16x8*>(
                 dO + q_base + (long)(q0 + row) * D + d0);
+            *reinterpret_cast<bf16x8*>(dOs + row * KSLOT * 8 + d0) = dov;
             #pragma unroll
-            for (int i = 0; i < 8; ++i) {
-                Qt[(d0 + i) * VROW + row] = qv[i];
+            for (int i = 0; i < 8; ++i)
                 dOt[(d0 + i) * VROW + row] = dov[i];
-            }
         }
     };
 
-    const int q_start = kb;    // causal: q tiles from the block's kv start
+    const int q_start = kvtile * KVB;   // causal: from the block's kv start
+    issue_loads(q_start);
     for (int q0 = q_start; q0 < S; q0 += CTILE) {
-        stage_tile(q0);
+        write_stage(q0);
+        if (q0 + CTILE < S) issue_loads(q0 + CTILE);
         __syncthreads();
 
-        bool rb_active[2];
-        #pragma unroll
-        for (int rb = 0; rb < 2; ++rb) {
-            const int kv0 = kb + rbid[rb] * 16;
-            rb_active[rb] = q0 + CTILE - 1 >= kv0;
-            if (!rb_active[rb]) continue;
-
+        const bool active = q0 + CTILE - 1 >= kv0;
+        if (active) {
             __builtin_amdgcn_s_setprio(1);
             floatx4 st_acc[4], dpt_acc[4];
             #pragma unroll
             for (int j = 0; j < 4; ++j) {
                 st_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
                 dpt_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
-                const long qrow = q0 + j * 16 + col16;
+                const int qrow = j * 16 + col16;
                 #pragma unroll
                 for (int c = 0; c < dchunks; ++c) {
                     const int d0 = c * 32 + k8 * 8;
+                    bf16x8 qf, dof;
+                    if (d0 < D) {
+                        qf = *reinterpret_cast<const bf16x8*>(
+                            Qs + qrow * KSLOT * 8 + d0);
+                        dof = *reinterpret_cast<const bf16x8*>(
+                            dOs + qrow * KSLOT * 8 + d0);
+                    } else {
+                        #pragma unroll
+                        for (int i = 0; i < 8; ++i) { qf[i] = 0; dof[i] = 0; }
+                    }
                     st_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        k_frag[rb][c], frag8<D>(Q + q_base, qrow, d0),
-                        st_acc[j], 0, 0, 0);
+                        k_frag[c], qf, st_acc[j], 0, 0, 0);
                     dpt_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        v_frag[rb][c], frag8<D>(dO + q_base, qrow, d0),
-                        dpt_acc[j], 0, 0, 0);
+                        v_frag[c], dof, dpt_acc[j], 0, 0, 0);
                 }
             }
             __builtin_amdgcn_s_setprio(0);
 
-            short* Srb = Sw + rb * 16 * VROW;
-            short* Prb = Pw + rb * 16 * VROW;
             #pragma unroll
             for (int j = 0; j < 4; ++j)
                 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
-                    const int kvrow = kb + rbid[rb] * 16 + k8 * 4 + r;
+                    const int kvrow = kv0 + k8 * 4 + r;
                     const int qcol = q0 + j * 16 + col16;
                     float p = (qcol < kvrow)
                                   ? 0.f
@@ -337,36 +370,30 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
                                            - LSE[row_base + qcol]);
                     float ds = scale * p
                                * (dpt_acc[j][r] - Delta[row_base + qcol]);
-                    Srb[(k8 * 4 + r) * VROW + j * 16 + col16] =
+                    Sw[(k8 * 4 + r) * VROW + j * 16 + col16] =
                         float_to_bf16_bits(ds);
-                    Prb[(k8 * 4 + r) * VROW + j * 16 + col16] =
+                    Pw[(k8 * 4 + r) * VROW + j * 16 + col16] =
                         float_to_bf16_bits(p);
                 }
-        }
 
-        asm volatile("" ::: "memory");   // wave-local dS^T / P^T publish
+            asm volatile("" ::: "memory");   // wave-local publish
 
-        #pragma unroll
-        for (int rb = 0; rb < 2; ++rb) {
-            if (!rb_active[rb]) continue;
-            const short* Srb = Sw + rb * 16 * VROW;
-            const short* Prb = Pw + rb * 16 * VROW;
             #pragma unroll
             for (int ks = 0; ks < 2; ++ks) {
                 bf16x8 dst_frag = *reinterpret_cast<const bf16x8*>(
-                    Srb + col16 * VROW + ks * 32 + k8 * 8);
+                    Sw + col16 * VROW + ks * 32 + k8 * 8);
                 bf16x8 pt_frag = *reinterpret_cast<const bf16x8*>(
-                    Prb + col16 * VROW + ks * 32 + k8 * 8);
+                    Pw + col16 * VROW + ks * 32 + k8 * 8);
                 #pragma unroll
                 for (int jd = 0; jd < djtiles; ++jd) {
                     bf16x8 qt_frag = *reinterpret_cast<const bf16x8*>(
                         Qt + (jd * 16 + col16) * VROW + ks * 32 + k8 * 8);
-                    dk_acc[rb][jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        dst_frag, qt_frag, dk_acc[rb][jd], 0, 0, 0);
+                    dk_acc[jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        dst_frag, qt_frag, dk_acc[jd], 0, 0, 0);
                     bf16x8 dot_frag = *reinterpret_cast<const bf16x8*>(
                         dOt + (jd * 16 + col16) * VROW + ks * 32 + k8 * 8);
-                    dv_acc[rb][jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        pt_frag, dot_frag, dv_acc[rb][jd], 0, 0, 0);
+                    dv_acc[jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        pt_frag, dot_frag, dv_acc[jd], 0, 0, 0);
                 }
             }
         }
@@ -374,18 +401,16 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     }
 
     #pragma unroll
-    for (int rb = 0; rb < 2; ++rb)
+    for (int r = 0; r < 4; ++r) {
+        const long kvrow = kv0 + k8 * 4 + r;
         #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            const long kvrow = kb + rbid[rb] * 16 + k8 * 4 + r;
-            #pragma unroll
-            for (int jd = 0; jd < djtiles; ++jd) {
-                dK[out_base + kvrow * D + jd * 16 + col16] =
-                    __float2bfloat16(dk_acc[rb][jd][r]);
-                dV[out_base + kvrow * D + jd * 16 + col16] =
-                    __float2bfloat16(dv_acc[rb][jd][r]);
-            }
+        for (int jd = 0; jd < djtiles; ++jd) {
+            dK[out_base + kvrow * D + jd * 16 + col16] =
+                __float2bfloat16(dk_acc[jd][r]);
+            dV[out_base + kvrow * D + jd * 16 + col16] =
+                __float2bfloat16(dv_acc[jd][r]);
         }
+    }
 }
 
 }  // namespace
@@ -408,12 +433,15 @@ std::vector<torch::Tensor> attn_bwd(
 
     auto stream = c10::hip::getCurrentHIPStream().stream();
     const int grid = (int)(B * H * (S / RBLK));
+    const int grid_dkv = (int)(B * H * (S / 64));
 
     #define LAUNCH_BWD(DD)                                                    \
         do {                                                                  \
             const int vrow = CTILE + VPAD;                                    \
-            const int lds_dq = (DD * vrow + 4 * 2 * 16 * vrow) * 2;           \
-            const int lds_dkv = (2 * DD * vrow + 2 * 8 * 16 * vrow) * 2;      \
+            const int lds_dq = (2 * CTILE * (DD / 8 + 1) * 8                  \
+                                + DD * vrow + 4 * 2 * 16 * vrow) * 2;         \
+            const int lds_dkv = (2 * CTILE * (DD / 8 + 1) * 8                 \
+                                 + 2 * DD * vrow + 2 * 4 * 16 * vrow) * 2;    \
             hipLaunchKernelGGL(attn_bwd_dq_kernel<DD>, dim3(grid),            \
                 dim3(THREADS), lds_dq, stream,                                \
                 reinterpret_cast<const bf16*>(qc.data_ptr()),                 \
@@ -423,7 +451,7 @@ std::vector<torch::Tensor> attn_bwd(
                 lsec.data_ptr<float>(), dc.data_ptr<float>(),                 \
                 reinterpret_cast<bf16*>(dq.data_ptr()),                       \
                 (int)B, (int)H, (int)Hkv, (int)S, (float)scale);              \
-            hipLaunchKernelGGL(attn_bwd_dkv_kernel<DD>, dim3(grid),           \
+            hipLaunchKernelGGL(attn_bwd_dkv_kernel<DD>, dim3(grid_dkv),       \
                 dim3(THREADS), lds_dkv, stream,                               \
                 reinterpret_cast<const bf16*>(qc.data_ptr()),                 \
                 reinterpret_cast<const bf16*>(kc.data_ptr()),                 \
