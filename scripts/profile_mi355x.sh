@@ -20,14 +20,14 @@ echo "== full-clock profile ($MODEL, bs=$BS) =="
 python -m metis_amd.profiler.profile_model --model "$MODEL" --bs "$BS" \
     --device-type MI355X --out profiles/mi355x --iters "$ITERS" --warmup 3
 
-echo "== clock-capped profile (MI355X_LC @ ${CAP_MHZ} MHz) =="
-if rocm-smi --setperfdeterminism "$CAP_MHZ" >/dev/null 2>&1; then
-    python -m metis_amd.profiler.profile_model --model "$MODEL" --bs "$BS" \
-        --device-type MI355X_LC --out profiles/mi355x_lc --iters "$ITERS" --warmup 3
-    rocm-smi --resetperfdeterminism >/dev/null 2>&1 || true
-else
-    echo "rocm-smi clock capping unavailable; skipping MI355X_LC profile"
-fi
+# The second (slower) device type is emulated by restricting the process
+# to half the CUs (rocm-smi clock/power caps are rejected on this pool;
+# HSA_CU_MASK gives a genuine compute-capacity asymmetry).
+echo "== half-CU profile (MI355X_LC via HSA_CU_MASK) =="
+HSA_CU_MASK=0:0-127 python -m metis_amd.profiler.profile_model \
+    --model "$MODEL" --bs "$BS" \
+    --device-type MI355X_LC --out "profiles/mi355x_lc/$MODEL" \
+    --iters "$ITERS" --warmup 3
 
 echo "profiles written:"
 ls profiles/mi355x profiles/mi355x_lc 2>/dev/null
