@@ -25,6 +25,7 @@ from metis_amd.ops import FusedAdamW
 from metis_amd.planner.volume import uniform_layer_split
 from metis_amd.runtime.comm import ParallelContext
 from metis_amd.runtime.grad_sync import GradBucketSync
+from metis_amd.runtime.trace import tracer_from_env
 
 
 class PlanRunner:
@@ -66,6 +67,7 @@ class PlanRunner:
             self.model.to(ctx.device)
         self.optimizer = FusedAdamW(self.model.parameters(), lr=lr)
         self.dtype = dtype
+        self.tracer = tracer_from_env(ctx.rank)
         # bucketed, overlapped DP gradient all-reduce
         self.grad_sync = None
         if ctx.dp > 1 and ctx.dp_group is not None:
@@ -89,10 +91,13 @@ class PlanRunner:
             if self.grad_sync is not None and mb == self.num_microbatches - 1:
                 self.grad_sync.arm()
             tokens, labels = self.synthetic_batch()
-            loss = self.model(tokens, labels=labels)
-            (loss / self.num_microbatches).backward()
+            with self.tracer.span("forward"):
+                loss = self.model(tokens, labels=labels)
+            with self.tracer.span("backward"):
+                (loss / self.num_microbatches).backward()
             losses.append(loss.detach())
         self._sync_and_step()
+        self.tracer.next_step()
         return float(torch.stack(losses).mean())
 
     # --- GPipe step (pp > 1) ----------------------------------------------
@@ -160,13 +165,15 @@ class PlanRunner:
 
     # --- gradient sync + optimizer ----------------------------------------
     def _sync_and_step(self) -> None:
-        if self.grad_sync is not None:
-            # hooks copied + all-reduced the final-microbatch grads,
-            # overlapped with backward; wait and average
-            self.grad_sync.finish()
-        else:
-            self.optimizer.gather_grads()
-        self.optimizer.step(pre_gathered=True)
+        with self.tracer.span("grad_sync"):
+            if self.grad_sync is not None:
+                # hooks copied + all-reduced the final-microbatch grads,
+                # overlapped with backward; wait and average
+                self.grad_sync.finish()
+            else:
+                self.optimizer.gather_grads()
+        with self.tracer.span("optimizer"):
+            self.optimizer.step(pre_gathered=True)
 
     def train_step(self) -> float:
         if self.ctx.pp == 1:
@@ -216,4 +223,5 @@ class PlanRunner:
             dist.barrier()
         if torch.cuda.is_available():
             torch.cuda.synchronize()
+        self.tracer.export()
         return (time.perf_counter() - t0) * 1000.0 / steps

@@ -51,3 +51,23 @@ def test_hooks_fill_flat_buffer_when_armed():
     got = opt.grad_flat[: expected.numel()]
     assert torch.allclose(got, expected)
     sync.finish()
+
+
+def test_step_tracer(tmp_path):
+    from metis_amd.runtime.trace import StepTracer
+
+    tracer = StepTracer(str(tmp_path / "trace.json"))
+    for _ in range(2):
+        with tracer.span("forward"):
+            pass
+        with tracer.span("optimizer"):
+            pass
+        tracer.next_step()
+    summary = tracer.summary()
+    assert set(summary) == {"forward", "optimizer"}
+    tracer.export()
+    import json as _json
+
+    doc = _json.loads((tmp_path / "trace.json").read_text())
+    assert len(doc["traceEvents"]) == 4
+    assert "summary_ms" in doc["metadata"]
