@@ -68,6 +68,7 @@ class PlanRunner:
         self.optimizer = FusedAdamW(self.model.parameters(), lr=lr)
         self.dtype = dtype
         self.tracer = tracer_from_env(ctx.rank)
+        self._data_gen = None
         # bucketed, overlapped DP gradient all-reduce
         self.grad_sync = None
         if ctx.dp > 1 and ctx.dp_group is not None:
@@ -75,10 +76,18 @@ class PlanRunner:
 
     # --- data -------------------------------------------------------------
     def synthetic_batch(self) -> Tuple[torch.Tensor, torch.Tensor]:
-        """Random tokens + next-token labels of the plan's microbatch shape."""
+        """Random tokens + next-token labels of the plan's microbatch shape.
+
+        Seeded per DP replica: all TP (and PP) ranks of one replica must
+        draw the SAME tokens, or tensor-parallel math silently diverges
+        across ranks; different DP replicas draw different data."""
         dev = self.ctx.device or torch.device("cpu")
+        if self._data_gen is None:
+            self._data_gen = torch.Generator(device=dev)
+            self._data_gen.manual_seed(12345 + self.ctx.dp_rank)
         tokens = torch.randint(
-            0, self.spec.vocab_size, (self.mbs, self.spec.seq_length), device=dev
+            0, self.spec.vocab_size, (self.mbs, self.spec.seq_length),
+            device=dev, generator=self._data_gen,
         )
         labels = torch.roll(tokens, -1, dims=1)
         return tokens, labels
@@ -132,12 +141,9 @@ class PlanRunner:
                 if ctx.is_first_stage:
                     pass  # pp == 1 handled elsewhere
                 else:
-                    # labels are generated on the last stage for synthetic data
-                    labels = torch.randint(
-                        0, self.spec.vocab_size,
-                        (self.mbs, self.spec.seq_length),
-                        device=ctx.device or "cpu",
-                    )
+                    # labels generated on the last stage (same per-replica
+                    # seeded stream as the first stage's tokens)
+                    _, labels = self.synthetic_batch()
                 out = self.model(x, labels=labels)
                 losses.append(out)
                 outputs.append(out)
