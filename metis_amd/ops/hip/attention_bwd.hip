@@ -287,11 +287,14 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
         dv_acc[jd] = floatx4{0.f, 0.f, 0.f, 0.f};
     }
 
-    // T14 staged Q (natural + transpose from one register set) plus a
-    // direct-staged dO pair; q-row-major lane walk
+    // T14 staged Q and dO (each register set feeds both the natural and
+    // transposed LDS image); at D=128 the dO set would spill past 256
+    // VGPRs on top of the dK+dV accumulators, so dO stages direct there.
     constexpr int chunks = CTILE * D / 8;
     constexpr int per_thread = (chunks + THREADS - 1) / THREADS;
+    constexpr bool STAGE_DO = (D <= 96);
     bf16x8 q_stage[per_thread];
+    bf16x8 do_stage[STAGE_DO ? per_thread : 1];
     auto issue_loads = [&](int q0) {
         #pragma unroll
         for (int u = 0; u < per_thread; ++u) {
@@ -299,6 +302,9 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
             if (c >= chunks) break;
             q_stage[u] = *reinterpret_cast<const bf16x8*>(
                 Q + q_base + (long)(q0 + c % CTILE) * D + (c / CTILE) * 8);
+            if constexpr (STAGE_DO)
+                do_stage[u] = *reinterpret_cast<const bf16x8*>(
+                    dO + q_base + (long)(q0 + c % CTILE) * D + (c / CTILE) * 8);
         }
     };
     auto write_stage = [&](int q0) {
@@ -312,8 +318,12 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
             #pragma unroll
             for (int i = 0; i < 8; ++i)
                 Qt[(d0 + i) * VROW + row] = q_stage[u][i];
-            bf16x8 dov = *reinterpret_cast<const bf16x8*>(
-                dO + q_base + (long)(q0 + row) * D + d0);
+            bf16x8 dov;
+            if constexpr (STAGE_DO)
+                dov = do_stage[u];
+            else
+                dov = *reinterpret_cast<const bf16x8*>(
+                    dO + q_base + (long)(q0 + row) * D + d0);
             *reinterpret_cast<bf16x8*>(dOs + row * KSLOT * 8 + d0) = dov;
             #pragma unroll
             for (int i = 0; i < 8; ++i)
