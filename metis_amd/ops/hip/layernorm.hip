@@ -82,6 +82,9 @@ __global__ void ln_fwd_kernel(
 
 // Backward pass 1: dx per row + per-block partial dgamma/dbeta slabs.
 // dx = rstd * (dy*g - mean(dy*g) - xhat * mean(dy*g*xhat))
+// dgamma/dbeta partials accumulate in LDS across the block's rows (each
+// thread owns fixed columns) and hit global memory once per block — the
+// per-row global read-modify-write was ~2x this kernel's traffic.
 __global__ void ln_bwd_kernel(
     const bf16x8* __restrict__ dy,
     const bf16x8* __restrict__ x,
@@ -94,11 +97,11 @@ __global__ void ln_bwd_kernel(
     int rows,
     int hv) {
     __shared__ float scratch[BLOCK / WAVE_SIZE];
+    extern __shared__ __attribute__((aligned(16))) float acc_lds[];  // [2][H]
 
     const int H = hv * VEC;
-    float* dg = dgamma_part + (long)blockIdx.x * H;
-    float* db = dbeta_part + (long)blockIdx.x * H;
-    // zero this block's slab
+    float* dg = acc_lds;
+    float* db = acc_lds + H;
     for (int i = threadIdx.x; i < H; i += BLOCK) {
         dg[i] = 0.f;
         db[i] = 0.f;
@@ -143,12 +146,20 @@ __global__ void ln_bwd_kernel(
                 float dyf = bf16_bits_to_float(dv[k]);
                 float xhat = (bf16_bits_to_float(xv[k]) - mean) * rstd;
                 o[k] = float_to_bf16_bits(rstd * (dyf * g - s1 - xhat * s2));
-                dg[i * VEC + k] += dyf * xhat;
-                db[i * VEC + k] += dyf;
+                // [k][i] layout: consecutive threads hit consecutive banks
+                dg[k * hv + i] += dyf * xhat;
+                db[k * hv + i] += dyf;
             }
             dxrow[i] = o;
         }
         __syncthreads();
+    }
+
+    // one global write of this block's partials ([k][i] -> column order)
+    for (int i = threadIdx.x; i < H; i += BLOCK) {
+        const int col_i = i / VEC, col_k = i % VEC;
+        dgamma_part[(long)blockIdx.x * H + i] = dg[col_k * hv + col_i];
+        dbeta_part[(long)blockIdx.x * H + i] = db[col_k * hv + col_i];
     }
 }
 
@@ -217,8 +228,9 @@ std::vector<torch::Tensor> layernorm_bwd(
     auto dbeta_part = torch::empty({grid, H}, f32);
     auto gamma_f = gamma.to(torch::kFloat32).contiguous();
 
+    TORCH_CHECK(H <= 8192, "layernorm bwd supports hidden size <= 8192");
     hipLaunchKernelGGL(
-        ln_bwd_kernel, dim3(grid), dim3(BLOCK), 0,
+        ln_bwd_kernel, dim3(grid), dim3(BLOCK), 2 * H * sizeof(float),
         c10::hip::getCurrentHIPStream().stream(),
         reinterpret_cast<const bf16x8*>(dyc.data_ptr()),
         reinterpret_cast<const bf16x8*>(x.data_ptr()),

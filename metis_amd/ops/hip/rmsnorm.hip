@@ -61,9 +61,10 @@ __global__ void rms_bwd_kernel(
     float* __restrict__ dgamma_part,   // [gridDim.x, H]
     int rows, int hv) {
     __shared__ float scratch[BLOCK / WAVE_SIZE];
+    extern __shared__ __attribute__((aligned(16))) float acc_lds[];  // [H]
 
     const int H = hv * VEC;
-    float* dg = dgamma_part + (long)blockIdx.x * H;
+    float* dg = acc_lds;   // per-block LDS accumulator (global write once)
     for (int i = threadIdx.x; i < H; i += BLOCK) dg[i] = 0.f;
 
     for (int row = blockIdx.x; row < rows; row += gridDim.x) {
@@ -95,12 +96,15 @@ __global__ void rms_bwd_kernel(
                 float dyf = bf16_bits_to_float(dv[k]);
                 float xhat = bf16_bits_to_float(xv[k]) * rstd;
                 o[k] = float_to_bf16_bits(rstd * (dyf * g) - rstd * xhat * s2);
-                dg[i * VEC + k] += dyf * xhat;
+                dg[k * hv + i] += dyf * xhat;   // [k][i]: bank-friendly
             }
             dxrow[i] = o;
         }
         __syncthreads();
     }
+
+    for (int i = threadIdx.x; i < H; i += BLOCK)
+        dgamma_part[(long)blockIdx.x * H + i] = dg[(i % VEC) * hv + i / VEC];
 }
 
 __global__ void rms_bwd_reduce_kernel(
@@ -150,7 +154,8 @@ std::vector<torch::Tensor> rmsnorm_bwd(
     auto dgamma_part = torch::empty({grid, H}, x.options().dtype(torch::kFloat32));
     auto gamma_f = gamma.to(torch::kFloat32).contiguous();
 
-    hipLaunchKernelGGL(rms_bwd_kernel, dim3(grid), dim3(BLOCK), 0,
+    TORCH_CHECK(H <= 16384, "rmsnorm bwd supports hidden size <= 16384");
+    hipLaunchKernelGGL(rms_bwd_kernel, dim3(grid), dim3(BLOCK), H * (int)sizeof(float),
         c10::hip::getCurrentHIPStream().stream(),
         reinterpret_cast<const bf16x8*>(dyc.data_ptr()),
         reinterpret_cast<const bf16x8*>(x.data_ptr()),
