@@ -24,6 +24,7 @@ import torch.nn.functional as F
 
 from metis_amd.ops import LayerNorm
 from metis_amd.ops.attention import flash_attention
+from metis_amd.ops.cross_entropy import cross_entropy
 
 
 @dataclass(frozen=True)
@@ -280,11 +281,11 @@ class GPTModel(nn.Module):
         return x
 
     def _loss(self, logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
-        flat = logits.float().view(-1, logits.size(-1))
         labels = labels.reshape(-1)
         if self.tp_group is not None and dist.get_world_size(self.tp_group) > 1:
+            flat = logits.float().view(-1, logits.size(-1))
             return _VocabParallelCrossEntropy.apply(flat, labels, self.tp_group)
-        return F.cross_entropy(flat, labels)
+        return cross_entropy(logits.view(-1, logits.size(-1)), labels)
 
     def layer_parameter_bytes(self) -> List[float]:
         """parameters_per_layer_bytes for the profile JSON (full model)."""

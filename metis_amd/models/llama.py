@@ -19,6 +19,7 @@ from metis_amd.models.gpt import (
     _VocabParallelCrossEntropy,
 )
 from metis_amd.ops.attention import flash_attention
+from metis_amd.ops.cross_entropy import cross_entropy
 from metis_amd.ops.norms import RMSNorm, apply_rope, swiglu
 
 
@@ -154,11 +155,11 @@ class LlamaModel(nn.Module):
             x = self.norm_final(x)
             logits = self.head(x, self.tp_group)
             if labels is not None:
-                flat = logits.float().view(-1, logits.size(-1))
                 labels = labels.reshape(-1)
                 if self.tp_group is not None and dist.get_world_size(self.tp_group) > 1:
+                    flat = logits.float().view(-1, logits.size(-1))
                     return _VocabParallelCrossEntropy.apply(flat, labels, self.tp_group)
-                return F.cross_entropy(flat, labels)
+                return cross_entropy(logits.view(-1, logits.size(-1)), labels)
             return logits
         return x
 

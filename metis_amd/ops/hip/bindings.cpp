@@ -32,6 +32,11 @@ torch::Tensor rope_apply(
 torch::Tensor swiglu_fwd(torch::Tensor a, torch::Tensor b);
 std::vector<torch::Tensor> swiglu_bwd(
     torch::Tensor dy, torch::Tensor a, torch::Tensor b);
+std::vector<torch::Tensor> cross_entropy_fwd(
+    torch::Tensor logits, torch::Tensor labels);
+torch::Tensor cross_entropy_bwd(
+    torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
+    torch::Tensor grad_scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("layernorm_fwd", &layernorm_fwd,
@@ -53,4 +58,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rope_apply", &rope_apply, "rotary embedding (fwd/bwd by flag)");
     m.def("swiglu_fwd", &swiglu_fwd, "fused silu(a)*b");
     m.def("swiglu_bwd", &swiglu_bwd, "fused SwiGLU backward");
+    m.def("cross_entropy_fwd", &cross_entropy_fwd,
+          "fused CE over bf16 logits -> (per-row loss, lse)");
+    m.def("cross_entropy_bwd", &cross_entropy_bwd,
+          "fused CE backward -> bf16 dlogits");
 }

@@ -1,0 +1,139 @@
+// Fused cross entropy over bf16 logits for gfx950.
+//
+// Replaces logits.float() + torch CE: the [tokens, vocab] fp32 copy is
+// never materialized (6.7 GB for the GPT-3 2.7B flagship step at gbs 16).
+// Forward: per-row two-pass max / sum-exp streaming the bf16 row
+// (vectorized 16 B/lane), emits per-row loss and LSE. Backward: one pass
+// writing bf16 dlogits = (softmax - onehot) * grad_scale.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int BLOCK = 256;
+constexpr int VEC = 8;
+
+__global__ void ce_fwd_kernel(
+    const bf16x8* __restrict__ logits,   // [N, V/8]
+    const long* __restrict__ labels,     // [N]
+    float* __restrict__ loss,            // [N]
+    float* __restrict__ lse,             // [N]
+    int rows, int vv) {                  // vv = V / 8
+    __shared__ float scratch[BLOCK / WAVE_SIZE];
+
+    for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+        const bf16x8* lrow = logits + (long)row * vv;
+
+        float m = -1e30f;
+        for (int i = threadIdx.x; i < vv; i += BLOCK) {
+            bf16x8 v = lrow[i];
+            #pragma unroll
+            for (int k = 0; k < VEC; ++k)
+                m = fmaxf(m, bf16_bits_to_float(v[k]));
+        }
+        // block max via the sum scratch (negate trick not needed: reduce manually)
+        m = wave_reduce_max(m);
+        const int wv = threadIdx.x / WAVE_SIZE;
+        if ((threadIdx.x & 63) == 0) scratch[wv] = m;
+        __syncthreads();
+        float bm = (threadIdx.x < BLOCK / WAVE_SIZE) ? scratch[threadIdx.x] : -1e30f;
+        bm = wave_reduce_max(bm);
+        if (threadIdx.x == 0) scratch[0] = bm;
+        __syncthreads();
+        m = scratch[0];
+        __syncthreads();
+
+        float sum = 0.f;
+        for (int i = threadIdx.x; i < vv; i += BLOCK) {
+            bf16x8 v = lrow[i];
+            #pragma unroll
+            for (int k = 0; k < VEC; ++k)
+                sum += __expf(bf16_bits_to_float(v[k]) - m);
+        }
+        sum = block_reduce_sum(sum, scratch);
+
+        if (threadIdx.x == 0) {
+            const long tgt = labels[row];
+            const float tl = bf16_bits_to_float(
+                reinterpret_cast<const short*>(lrow)[tgt]);
+            const float l = m + __logf(sum);
+            lse[row] = l;
+            loss[row] = l - tl;
+        }
+        __syncthreads();
+    }
+}
+
+__global__ void ce_bwd_kernel(
+    const bf16x8* __restrict__ logits,
+    const long* __restrict__ labels,
+    const float* __restrict__ lse,
+    const float* __restrict__ grad_scale,  // dloss/dmean * 1/N, scalar tensor
+    bf16x8* __restrict__ dlogits,
+    int rows, int vv) {
+    const float gs = *grad_scale;
+    for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+        const bf16x8* lrow = logits + (long)row * vv;
+        bf16x8* drow = dlogits + (long)row * vv;
+        const float l = lse[row];
+        const long tgt = labels[row];
+        for (int i = threadIdx.x; i < vv; i += BLOCK) {
+            bf16x8 v = lrow[i];
+            bf16x8 o;
+            #pragma unroll
+            for (int k = 0; k < VEC; ++k) {
+                float p = __expf(bf16_bits_to_float(v[k]) - l);
+                if ((long)i * VEC + k == tgt) p -= 1.f;
+                o[k] = float_to_bf16_bits(p * gs);
+            }
+            drow[i] = o;
+        }
+    }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> cross_entropy_fwd(
+    torch::Tensor logits, torch::Tensor labels) {
+    TORCH_CHECK(logits.is_cuda() && logits.dtype() == torch::kBFloat16);
+    TORCH_CHECK(logits.dim() == 2 && logits.size(1) % 8 == 0,
+                "logits must be [N, V] with V % 8 == 0");
+    TORCH_CHECK(labels.dtype() == torch::kInt64);
+    auto lc = logits.contiguous();
+    const long N = logits.size(0), V = logits.size(1);
+
+    auto f32 = logits.options().dtype(torch::kFloat32);
+    auto loss = torch::empty({N}, f32);
+    auto lse = torch::empty({N}, f32);
+    const int grid = (int)std::min<long>(N, 2048);
+    hipLaunchKernelGGL(ce_fwd_kernel, dim3(grid), dim3(BLOCK), 0,
+        c10::hip::getCurrentHIPStream().stream(),
+        reinterpret_cast<const bf16x8*>(lc.data_ptr()),
+        labels.contiguous().data_ptr<long>(),
+        loss.data_ptr<float>(), lse.data_ptr<float>(),
+        (int)N, (int)(V / 8));
+    HIP_CHECK_LAST();
+    return {loss, lse};
+}
+
+torch::Tensor cross_entropy_bwd(
+    torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
+    torch::Tensor grad_scale) {
+    auto lc = logits.contiguous();
+    const long N = logits.size(0), V = logits.size(1);
+    auto dlogits = torch::empty_like(lc);
+    const int grid = (int)std::min<long>(N, 2048);
+    hipLaunchKernelGGL(ce_bwd_kernel, dim3(grid), dim3(BLOCK), 0,
+        c10::hip::getCurrentHIPStream().stream(),
+        reinterpret_cast<const bf16x8*>(lc.data_ptr()),
+        labels.contiguous().data_ptr<long>(),
+        lse.data_ptr<float>(),
+        grad_scale.contiguous().data_ptr<float>(),
+        reinterpret_cast<bf16x8*>(dlogits.data_ptr()),
+        (int)N, (int)(V / 8));
+    HIP_CHECK_LAST();
+    return dlogits;
+}

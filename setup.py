@@ -27,6 +27,7 @@ sources = [
     os.path.join(HIP_DIR, "rmsnorm.hip"),
     os.path.join(HIP_DIR, "rope.hip"),
     os.path.join(HIP_DIR, "swiglu.hip"),
+    os.path.join(HIP_DIR, "cross_entropy.hip"),
 ]
 sources = [s for s in sources if os.path.exists(s)]
 

@@ -118,3 +118,24 @@ def test_model_step_gpu():
         opt.step()
         losses.append(float(loss.detach()))
     assert losses[-1] < losses[0], losses
+
+
+def test_fused_cross_entropy_matches_fp32():
+    from metis_amd.ops.cross_entropy import cross_entropy
+
+    torch.manual_seed(5)
+    n, v = 512, 51200
+    logits = torch.randn(n, v, device="cuda", dtype=torch.bfloat16,
+                         requires_grad=True)
+    labels = torch.randint(0, v, (n,), device="cuda")
+
+    loss = cross_entropy(logits, labels)
+    loss.backward()
+
+    lf = logits.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(lf, labels)
+    ref.backward()
+
+    assert abs(float(loss) - float(ref)) / float(ref) < 2e-3
+    err = (logits.grad.float() - lf.grad).abs().max()
+    assert err < 1e-4, err
