@@ -7,7 +7,7 @@ quirk Q9): everything below is an explicit dataclass.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 

@@ -11,7 +11,6 @@ from typing import List, Optional
 import torch
 import torch.distributed as dist
 import torch.nn as nn
-import torch.nn.functional as F
 
 from metis_amd.models.gpt import (
     ColumnParallelLinear,

@@ -10,8 +10,6 @@ Dispatch policy:
 
 from __future__ import annotations
 
-import torch
-
 _EXT = None
 _EXT_ERR: str = ""
 

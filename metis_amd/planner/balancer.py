@@ -11,7 +11,6 @@ The heuristics' magic constants are kept and named:
 from __future__ import annotations
 
 import copy
-import math
 from typing import Dict, List, Optional, Sequence, Tuple
 
 from metis_amd.cluster import ClusterSpec, DeviceSpec

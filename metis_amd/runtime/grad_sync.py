@@ -13,7 +13,7 @@ the final microbatch's backward (DDP no_sync pattern).
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.distributed as dist
