@@ -63,7 +63,6 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_kernel(
     const int tile_n = wg % ntile_n;
 
     const int kv = K / 8;            // 16B chunks per row
-    const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
     const int wm = wave >> 1;        // wave row (0..1)
     const int wn = wave & 1;         // wave col (0..1)
@@ -79,21 +78,36 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_bf16_kernel(
             smem + buf * 2 * (BM + BN) * BK + 2 * BM * BK);
     };
 
-    // staging: 256 threads x 16 B: A tile = 128x64 bf16 = 1024 chunks ->
-    // 4 chunks/thread; same for B. Thread t stages rows [t/8]x4, slot t%8.
-    const int st_row = (threadIdx.x >> 3) * 4;  // 4 consecutive rows
-    const int st_slot = threadIdx.x & 7;
+    // staging: global_load_lds DMA, 16 B per lane (ladder step 3 — the
+    // compiler never auto-emits it; width 4 -> 16 alone is +67% there).
+    // The LDS image is lane-linear (chunk p = row*8 + slot), so the XOR
+    // swizzle moves to the SOURCE address: position (row, s) receives
+    // source slot s ^ (row & 7) — the same involution the reads apply
+    // (guide rule 21: both sides or neither).
+    typedef __attribute__((address_space(1))) const unsigned int glds_src_t;
+    typedef __attribute__((address_space(3))) unsigned int glds_dst_t;
+    const int lane = threadIdx.x & 63;
+    const int wave4 = threadIdx.x >> 6;
 
     auto stage = [&](int buf, int k0) {
         const long a_base = (long)tile_m * BM;
         const long b_base = (long)tile_n * BN;
         #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            int row = st_row + r;
-            bf16x8 av = A[(a_base + row) * kv + k0 / 8 + st_slot];
-            sA(buf)[swz_slot(row, st_slot)] = av;
-            bf16x8 bv = B[(b_base + row) * kv + k0 / 8 + st_slot];
-            sB(buf)[swz_slot(row, st_slot)] = bv;
+        for (int j = 0; j < 4; ++j) {
+            const int chunkbase = (j * 4 + wave4) * 64;   // wave-uniform
+            const int p = chunkbase + lane;
+            const int row = p >> 3;
+            const int src_slot = (p & 7) ^ (row & 7);
+            const bf16x8* asrc = A + (a_base + row) * kv + k0 / 8 + src_slot;
+            __builtin_amdgcn_global_load_lds(
+                (glds_src_t*)asrc,
+                (glds_dst_t*)(reinterpret_cast<char*>(sA(buf)) + chunkbase * 16),
+                16, 0, 0);
+            const bf16x8* bsrc = B + (b_base + row) * kv + k0 / 8 + src_slot;
+            __builtin_amdgcn_global_load_lds(
+                (glds_src_t*)bsrc,
+                (glds_dst_t*)(reinterpret_cast<char*>(sB(buf)) + chunkbase * 16),
+                16, 0, 0);
         }
     };
 
