@@ -223,7 +223,7 @@ std::vector<torch::Tensor> layernorm_bwd(
 
     auto dx = torch::empty_like(x);
     auto f32 = x.options().dtype(torch::kFloat32);
-    const int grid = (int)std::min<long>(rows, 512);
+    const int grid = (int)std::min<long>(rows, 128);
     auto dgamma_part = torch::empty({grid, H}, f32);
     auto dbeta_part = torch::empty({grid, H}, f32);
     auto gamma_f = gamma.to(torch::kFloat32).contiguous();

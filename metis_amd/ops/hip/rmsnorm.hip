@@ -150,7 +150,7 @@ std::vector<torch::Tensor> rmsnorm_bwd(
     const long rows = x.numel() / H;
 
     auto dx = torch::empty_like(x);
-    const int grid = (int)std::min<long>(rows, 512);
+    const int grid = (int)std::min<long>(rows, 128);
     auto dgamma_part = torch::empty({grid, H}, x.options().dtype(torch::kFloat32));
     auto gamma_f = gamma.to(torch::kFloat32).contiguous();
 

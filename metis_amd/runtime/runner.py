@@ -69,10 +69,11 @@ class PlanRunner:
         self.dtype = dtype
         self.tracer = tracer_from_env(ctx.rank)
         self._data_gen = None
-        # bucketed, overlapped DP gradient all-reduce
-        self.grad_sync = None
-        if ctx.dp > 1 and ctx.dp_group is not None:
-            self.grad_sync = GradBucketSync(self.optimizer, ctx.dp_group, ctx.dp)
+        # bucketed DP gradient sync; at dp == 1 it still runs (group None):
+        # the hooks copy grads into the optimizer's flat buffer DURING
+        # backward instead of a serial per-parameter gather afterwards
+        self.grad_sync = GradBucketSync(
+            self.optimizer, ctx.dp_group if ctx.dp > 1 else None, ctx.dp)
 
     # --- data -------------------------------------------------------------
     def synthetic_batch(self) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -172,12 +173,9 @@ class PlanRunner:
     # --- gradient sync + optimizer ----------------------------------------
     def _sync_and_step(self) -> None:
         with self.tracer.span("grad_sync"):
-            if self.grad_sync is not None:
-                # hooks copied + all-reduced the final-microbatch grads,
-                # overlapped with backward; wait and average
-                self.grad_sync.finish()
-            else:
-                self.optimizer.gather_grads()
+            # hooks copied (+ all-reduced when dp > 1) the final-microbatch
+            # grads, overlapped with backward; wait and average
+            self.grad_sync.finish()
         with self.tracer.span("optimizer"):
             self.optimizer.step(pre_gathered=True)
 
