@@ -25,8 +25,13 @@ namespace {
 
 constexpr int THREADS = 256;
 constexpr int QBLK = 128;     // q rows per block (2 x 16 per wave)
-constexpr int KVBLK = 64;     // kv columns per tile
 constexpr int VPAD = 8;       // Vt row padding (bf16 elements)
+
+// KV tile width: 128 for D <= 80 (the K/V^T/P LDS images still fit two
+// blocks per CU), 64 above. Wider tiles amortize the two barriers per
+// tile over twice the MFMA work.
+template <int D>
+constexpr int kvblk_for() { return D <= 80 ? 128 : 64; }
 
 template <int D>
 __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
@@ -42,6 +47,8 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
     const int col16 = lane & 15;
     const int k8 = lane >> 4;
 
+    constexpr int KVBLK = kvblk_for<D>();    // kv columns per tile
+    constexpr int JSUB = KVBLK / 16;         // 16-col S subtiles per tile
     constexpr int dchunks = (D + 31) / 32;   // 32-wide K-dim chunks of D
     constexpr int djtiles = D / 16;          // 16-wide output column tiles
     constexpr int KSLOT = D / 8 + 1;         // K LDS slots per row (padded)
@@ -153,7 +160,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
 
         // ---- per row block: S = scale * Q K^T, online softmax, P -> LDS -
         bool rb_active[2];
-        float p[2][4][4];   // [rb][j][r]
+        float p[2][JSUB][4];   // [rb][j][r]
         #pragma unroll
         for (int rb = 0; rb < 2; ++rb) {
             const int q0 = qb + rbid[rb] * 16;
@@ -162,7 +169,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
 
             __builtin_amdgcn_s_setprio(1);   // T5: favor the MFMA cluster
             #pragma unroll
-            for (int j = 0; j < 4; ++j) {       // four 16-col subtiles
+            for (int j = 0; j < JSUB; ++j) {    // 16-col subtiles
                 floatx4 s_acc = floatx4{0.f, 0.f, 0.f, 0.f};
                 const int kvrow = j * 16 + col16;
                 #pragma unroll
@@ -193,7 +200,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
             for (int r = 0; r < 4; ++r) {
                 float tile_max = -1e30f;
                 #pragma unroll
-                for (int j = 0; j < 4; ++j)
+                for (int j = 0; j < JSUB; ++j)
                     tile_max = fmaxf(tile_max, p[rb][j][r]);
                 #pragma unroll
                 for (int off = 8; off > 0; off >>= 1)
@@ -203,7 +210,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
                 const float alpha = __expf(m_run[rb][r] - m_new);
                 float row_sum = 0.f;
                 #pragma unroll
-                for (int j = 0; j < 4; ++j) {
+                for (int j = 0; j < JSUB; ++j) {
                     p[rb][j][r] = __expf(p[rb][j][r] - m_new);
                     row_sum += p[rb][j][r];
                 }
@@ -220,7 +227,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
 
             short* Prb = Pw + rb * 16 * VROW;
             #pragma unroll
-            for (int j = 0; j < 4; ++j)
+            for (int j = 0; j < JSUB; ++j)
                 #pragma unroll
                 for (int r = 0; r < 4; ++r)
                     Prb[(k8 * 4 + r) * VROW + j * 16 + col16] =
@@ -238,7 +245,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
             if (!rb_active[rb]) continue;
             const short* Prb = Pw + rb * 16 * VROW;
             #pragma unroll
-            for (int ks = 0; ks < 2; ++ks) {    // two 32-wide kv chunks
+            for (int ks = 0; ks < KVBLK / 32; ++ks) {   // 32-wide kv chunks
                 bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(
                     Prb + col16 * VROW + ks * 32 + k8 * 8);
                 #pragma unroll
@@ -291,8 +298,9 @@ std::vector<torch::Tensor> attn_fwd(
     auto stream = c10::hip::getCurrentHIPStream().stream();
     #define LAUNCH_D(DD)                                                      \
         do {                                                                  \
-            const int lds = (KVBLK * (DD / 8 + 1) * 8 + DD * (KVBLK + VPAD)   \
-                             + 4 * 2 * 16 * (KVBLK + VPAD)) * 2;              \
+            const int kvb = kvblk_for<DD>();                                  \
+            const int lds = (kvb * (DD / 8 + 1) * 8 + DD * (kvb + VPAD)       \
+                             + 4 * 2 * 16 * (kvb + VPAD)) * 2;                \
             hipLaunchKernelGGL(attn_fwd_kernel<DD>, dim3(grid),               \
                 dim3(THREADS), lds, stream,                                   \
                 reinterpret_cast<const bf16*>(qc.data_ptr()),                 \
