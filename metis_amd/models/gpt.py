@@ -25,6 +25,7 @@ import torch.nn.functional as F
 from metis_amd.ops import LayerNorm
 from metis_amd.ops.attention import flash_attention
 from metis_amd.ops.cross_entropy import cross_entropy
+from metis_amd.ops.relayout import heads_merge, qkv_split_transpose
 
 
 @dataclass(frozen=True)
@@ -194,18 +195,15 @@ class GPTBlock(nn.Module):
         self.fc2 = RowParallelLinear(spec.ffn, h, tp, dtype)
 
     def forward(self, x: torch.Tensor, tp_group) -> torch.Tensor:
-        b, s, _ = x.shape
         residual = x
         y = self.ln_attn(x)
+        # per-rank qkv layout: [q heads | k heads | v heads] blocks
         qkv = self.qkv(y, tp_group)
-        qkv = qkv.view(b, s, self.heads_per_rank, 3 * self.head_dim)
-        q, k, v = qkv.chunk(3, dim=-1)
-        q = q.transpose(1, 2).contiguous()  # [b, heads, s, d]
-        k = k.transpose(1, 2).contiguous()
-        v = v.transpose(1, 2).contiguous()
+        q, k, v = qkv_split_transpose(
+            qkv, self.heads_per_rank, self.heads_per_rank, self.head_dim
+        )
         attn = flash_attention(q, k, v, causal=True)
-        attn = attn.transpose(1, 2).reshape(b, s, -1)
-        x = residual + self.proj(attn, tp_group)
+        x = residual + self.proj(heads_merge(attn), tp_group)
 
         residual = x
         y = self.ln_mlp(x)

@@ -20,6 +20,7 @@ from metis_amd.models.gpt import (
 from metis_amd.ops.attention import flash_attention
 from metis_amd.ops.cross_entropy import cross_entropy
 from metis_amd.ops.norms import RMSNorm, apply_rope, swiglu
+from metis_amd.ops.relayout import heads_merge, qkv_split_transpose
 
 
 @dataclass(frozen=True)
@@ -83,23 +84,16 @@ class LlamaBlock(nn.Module):
         self.ffn_per_rank = spec.ffn_hidden_size // tp
 
     def forward(self, x: torch.Tensor, tp_group) -> torch.Tensor:
-        b, s, _ = x.shape
         hq, hkv, d = self.heads_per_rank, self.kv_heads_per_rank, self.head_dim
 
         residual = x
         y = self.norm_attn(x)
         qkv = self.qkv(y, tp_group)
-        q, k, v = qkv.split(
-            [hq * d, hkv * d, hkv * d], dim=-1
-        )
-        q = q.view(b, s, hq, d).transpose(1, 2).contiguous()
-        k = k.view(b, s, hkv, d).transpose(1, 2).contiguous()
-        v = v.view(b, s, hkv, d).transpose(1, 2).contiguous()
+        q, k, v = qkv_split_transpose(qkv, hq, hkv, d)
         q = apply_rope(q, self.rope_base)
         k = apply_rope(k, self.rope_base)
         attn = flash_attention(q, k, v, causal=True)
-        attn = attn.transpose(1, 2).reshape(b, s, hq * d)
-        x = residual + self.proj(attn, tp_group)
+        x = residual + self.proj(heads_merge(attn), tp_group)
 
         residual = x
         y = self.norm_mlp(x)

@@ -37,6 +37,12 @@ std::vector<torch::Tensor> cross_entropy_fwd(
 torch::Tensor cross_entropy_bwd(
     torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
     torch::Tensor grad_scale);
+std::vector<torch::Tensor> qkv_split_transpose(
+    torch::Tensor qkv, long nq, long nkv, long head_dim);
+torch::Tensor qkv_split_transpose_bwd(
+    torch::Tensor dq, torch::Tensor dk, torch::Tensor dv, long head_dim);
+torch::Tensor heads_merge(torch::Tensor x);
+torch::Tensor heads_unmerge(torch::Tensor y, long H);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("layernorm_fwd", &layernorm_fwd,
@@ -62,4 +68,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused CE over bf16 logits -> (per-row loss, lse)");
     m.def("cross_entropy_bwd", &cross_entropy_bwd,
           "fused CE backward -> bf16 dlogits");
+    m.def("qkv_split_transpose", &qkv_split_transpose,
+          "[B,S,(nq+2nkv)D] -> q/k/v [B,h,S,D] in one pass");
+    m.def("qkv_split_transpose_bwd", &qkv_split_transpose_bwd,
+          "dq/dk/dv -> fused dqkv layout");
+    m.def("heads_merge", &heads_merge, "[B,H,S,D] -> [B,S,H*D]");
+    m.def("heads_unmerge", &heads_unmerge, "[B,S,H*D] -> [B,H,S,D]");
 }

@@ -1,0 +1,65 @@
+"""Fused head relayouts around attention (gfx950 kernels, CPU fallback).
+
+One pass per direction instead of eager chunk/transpose/.contiguous()
+(six passes per transformer block).
+"""
+
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+from metis_amd import ops as _ops
+
+
+class _QKVSplitTranspose(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, qkv, nq, nkv, head_dim):
+        ext = _ops.require_extension()
+        ctx.head_dim = head_dim
+        q, k, v = ext.qkv_split_transpose(qkv, nq, nkv, head_dim)
+        return q, k, v
+
+    @staticmethod
+    def backward(ctx, dq, dk, dv):
+        ext = _ops.require_extension()
+        dqkv = ext.qkv_split_transpose_bwd(
+            dq.contiguous(), dk.contiguous(), dv.contiguous(), ctx.head_dim
+        )
+        return dqkv, None, None, None
+
+
+class _HeadsMerge(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ext = _ops.require_extension()
+        ctx.heads = x.size(1)
+        return ext.heads_merge(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _ops.require_extension()
+        return ext.heads_unmerge(dy.contiguous(), ctx.heads)
+
+
+def qkv_split_transpose(
+    qkv: torch.Tensor, nq: int, nkv: int, head_dim: int
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """qkv [B, S, (nq+2*nkv)*D] -> q [B,nq,S,D], k/v [B,nkv,S,D]."""
+    if qkv.is_cuda and qkv.dtype == torch.bfloat16 and head_dim % 8 == 0:
+        return _QKVSplitTranspose.apply(qkv, nq, nkv, head_dim)
+    b, s, _ = qkv.shape
+    q, k, v = qkv.split([nq * head_dim, nkv * head_dim, nkv * head_dim], -1)
+    q = q.view(b, s, nq, head_dim).transpose(1, 2).contiguous()
+    k = k.view(b, s, nkv, head_dim).transpose(1, 2).contiguous()
+    v = v.view(b, s, nkv, head_dim).transpose(1, 2).contiguous()
+    return q, k, v
+
+
+def heads_merge(x: torch.Tensor) -> torch.Tensor:
+    """[B, H, S, D] -> [B, S, H*D]."""
+    if x.is_cuda and x.dtype == torch.bfloat16 and x.size(-1) % 8 == 0:
+        return _HeadsMerge.apply(x)
+    b, h, s, d = x.shape
+    return x.transpose(1, 2).reshape(b, s, h * d)

@@ -28,6 +28,7 @@ sources = [
     os.path.join(HIP_DIR, "rope.hip"),
     os.path.join(HIP_DIR, "swiglu.hip"),
     os.path.join(HIP_DIR, "cross_entropy.hip"),
+    os.path.join(HIP_DIR, "relayout.hip"),
 ]
 sources = [s for s in sources if os.path.exists(s)]
 

@@ -139,3 +139,36 @@ def test_fused_cross_entropy_matches_fp32():
     assert abs(float(loss) - float(ref)) / float(ref) < 2e-3
     err = (logits.grad.float() - lf.grad).abs().max()
     assert err < 1e-4, err
+
+
+def test_relayout_kernels_match_eager():
+    from metis_amd.ops.relayout import heads_merge, qkv_split_transpose
+
+    torch.manual_seed(6)
+    b, s, nq, nkv, d = 2, 128, 8, 2, 64
+    qkv = torch.randn(b, s, (nq + 2 * nkv) * d, device="cuda",
+                      dtype=torch.bfloat16, requires_grad=True)
+    q, k, v = qkv_split_transpose(qkv, nq, nkv, d)
+
+    rq, rk, rv = qkv.split([nq * d, nkv * d, nkv * d], -1)
+    assert torch.equal(q, rq.view(b, s, nq, d).transpose(1, 2).contiguous())
+    assert torch.equal(k, rk.view(b, s, nkv, d).transpose(1, 2).contiguous())
+    assert torch.equal(v, rv.view(b, s, nkv, d).transpose(1, 2).contiguous())
+
+    # backward: gradients land back in the fused layout exactly
+    gq, gk, gv = torch.randn_like(q), torch.randn_like(k), torch.randn_like(v)
+    torch.autograd.backward([q, k, v], [gq, gk, gv])
+    ref = torch.cat([
+        gq.transpose(1, 2).reshape(b, s, -1),
+        gk.transpose(1, 2).reshape(b, s, -1),
+        gv.transpose(1, 2).reshape(b, s, -1),
+    ], dim=-1)
+    assert torch.equal(qkv.grad, ref)
+
+    x = torch.randn(b, nq, s, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = heads_merge(x)
+    assert torch.equal(y, x.transpose(1, 2).reshape(b, s, -1))
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    assert torch.equal(x.grad, gy.view(b, s, nq, d).transpose(1, 2).contiguous())

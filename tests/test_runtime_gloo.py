@@ -32,7 +32,17 @@ def _shard_from_full(full: GPTModel, shard: GPTModel, r: int, tp: int) -> None:
             for name in ("ln_attn", "ln_mlp"):
                 getattr(sb, name).weight.copy_(getattr(fb, name).weight)
                 getattr(sb, name).bias.copy_(getattr(fb, name).bias)
-            for name in ("qkv", "fc1"):  # column-parallel: shard output rows
+            # qkv uses block layout [q heads | k heads | v heads]: rank r's
+            # shard is the r-th head-slice of EACH block
+            h = fb.qkv.weight.size(0) // 3
+            hp = h // tp
+            rows = torch.cat([
+                torch.arange(blk * h + r * hp, blk * h + (r + 1) * hp)
+                for blk in range(3)
+            ])
+            sb.qkv.weight.copy_(fb.qkv.weight[rows])
+            sb.qkv.bias.copy_(fb.qkv.bias[rows])
+            for name in ("fc1",):  # plain column-parallel: shard output rows
                 fw, sw = getattr(fb, name), getattr(sb, name)
                 opr = sw.out_per_rank
                 sw.weight.copy_(fw.weight[r * opr:(r + 1) * opr])
