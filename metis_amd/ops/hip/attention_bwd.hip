@@ -264,10 +264,15 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
 
     const int kv0 = kvtile * KVB + wave * 16;   // this wave's kv rows
 
+    // At D <= 96 the natural Q/dO tiles fit in LDS next to the transposed
+    // ones at 2 blocks/CU; at D = 128 that allocation would drop to 1
+    // block/CU, so the natural fragments read straight from global (L2).
+    constexpr bool STAGE_NATURAL = (D <= 96);
+    constexpr int NATSZ = STAGE_NATURAL ? CTILE * KSLOT * 8 : 0;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     short* Qs = reinterpret_cast<short*>(smem);                    // [CTILE][KSLOT*8]
-    short* dOs = Qs + CTILE * KSLOT * 8;                           // [CTILE][KSLOT*8]
-    short* Qt = dOs + CTILE * KSLOT * 8;                           // [D][VROW]
+    short* dOs = Qs + NATSZ;                                       // [CTILE][KSLOT*8]
+    short* Qt = dOs + NATSZ;                                       // [D][VROW]
     short* dOt = Qt + D * VROW;                                    // [D][VROW]
     short* Sw = dOt + D * VROW + wave * 16 * VROW;                 // [16][VROW]
     short* Pw = dOt + D * VROW + 4 * 16 * VROW + wave * 16 * VROW; // [16][VROW]
@@ -314,7 +319,8 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
             if (c >= chunks) break;
             const int row = c % CTILE;
             const int d0 = (c / CTILE) * 8;
-            *reinterpret_cast<bf16x8*>(Qs + row * KSLOT * 8 + d0) = q_stage[u];
+            if constexpr (STAGE_NATURAL)
+                *reinterpret_cast<bf16x8*>(Qs + row * KSLOT * 8 + d0) = q_stage[u];
             #pragma unroll
             for (int i = 0; i < 8; ++i)
                 Qt[(d0 + i) * VROW + row] = q_stage[u][i];
@@ -324,7 +330,8 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
             else
                 dov = *reinterpret_cast<const bf16x8*>(
                     dO + q_base + (long)(q0 + row) * D + d0);
-            *reinterpret_cast<bf16x8*>(dOs + row * KSLOT * 8 + d0) = dov;
+            if constexpr (STAGE_NATURAL)
+                *reinterpret_cast<bf16x8*>(dOs + row * KSLOT * 8 + d0) = dov;
             #pragma unroll
             for (int i = 0; i < 8; ++i)
                 dOt[(d0 + i) * VROW + row] = dov[i];
@@ -351,14 +358,19 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
                 for (int c = 0; c < dchunks; ++c) {
                     const int d0 = c * 32 + k8 * 8;
                     bf16x8 qf, dof;
-                    if (d0 < D) {
-                        qf = *reinterpret_cast<const bf16x8*>(
-                            Qs + qrow * KSLOT * 8 + d0);
-                        dof = *reinterpret_cast<const bf16x8*>(
-                            dOs + qrow * KSLOT * 8 + d0);
+                    if constexpr (STAGE_NATURAL) {
+                        if (d0 < D) {
+                            qf = *reinterpret_cast<const bf16x8*>(
+                                Qs + qrow * KSLOT * 8 + d0);
+                            dof = *reinterpret_cast<const bf16x8*>(
+                                dOs + qrow * KSLOT * 8 + d0);
+                        } else {
+                            #pragma unroll
+                            for (int i = 0; i < 8; ++i) { qf[i] = 0; dof[i] = 0; }
+                        }
                     } else {
-                        #pragma unroll
-                        for (int i = 0; i < 8; ++i) { qf[i] = 0; dof[i] = 0; }
+                        qf = frag8<D>(Q + q_base, q0 + qrow, d0);
+                        dof = frag8<D>(dO + q_base, q0 + qrow, d0);
                     }
                     st_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         k_frag[c], qf, st_acc[j], 0, 0, 0);
@@ -450,7 +462,8 @@ std::vector<torch::Tensor> attn_bwd(
             const int vrow = CTILE + VPAD;                                    \
             const int lds_dq = (2 * CTILE * (DD / 8 + 1) * 8                  \
                                 + DD * vrow + 4 * 2 * 16 * vrow) * 2;         \
-            const int lds_dkv = (2 * CTILE * (DD / 8 + 1) * 8                 \
+            const int nat = DD <= 96 ? 2 * CTILE * (DD / 8 + 1) * 8 : 0;      \
+            const int lds_dkv = (nat                                          \
                                  + 2 * DD * vrow + 2 * 4 * 16 * vrow) * 2;    \
             hipLaunchKernelGGL(attn_bwd_dq_kernel<DD>, dim3(grid),            \
                 dim3(THREADS), lds_dq, stream,                                \
