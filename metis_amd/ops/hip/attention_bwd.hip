@@ -267,7 +267,10 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     // At D <= 96 the natural Q/dO tiles fit in LDS next to the transposed
     // ones at 2 blocks/CU; at D = 128 that allocation would drop to 1
     // block/CU, so the natural fragments read straight from global (L2).
-    constexpr bool STAGE_NATURAL = (D <= 96);
+    // Natural Q/dO tiles stay in LDS at every D: the D=128 allocation
+    // drops to 1 block/CU but measures 118 TF/s-eq vs 89 with 2 blocks
+    // reading natural fragments from L2 (A/B on B8 H32 S2048 D128).
+    constexpr bool STAGE_NATURAL = true;
     constexpr int NATSZ = STAGE_NATURAL ? CTILE * KSLOT * 8 : 0;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     short* Qs = reinterpret_cast<short*>(smem);                    // [CTILE][KSLOT*8]
@@ -462,7 +465,7 @@ std::vector<torch::Tensor> attn_bwd(
             const int vrow = CTILE + VPAD;                                    \
             const int lds_dq = (2 * CTILE * (DD / 8 + 1) * 8                  \
                                 + DD * vrow + 4 * 2 * 16 * vrow) * 2;         \
-            const int nat = DD <= 96 ? 2 * CTILE * (DD / 8 + 1) * 8 : 0;      \
+            const int nat = 2 * CTILE * (DD / 8 + 1) * 8;                     \
             const int lds_dkv = (nat                                          \
                                  + 2 * DD * vrow + 2 * 4 * 16 * vrow) * 2;    \
             hipLaunchKernelGGL(attn_bwd_dq_kernel<DD>, dim3(grid),            \
