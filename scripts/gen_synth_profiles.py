@@ -42,9 +42,9 @@ def layer_memory(tp: int, bs: int):
     return [m_embed] + [m_block] * (NUM_LAYERS - 2) + [m_head]
 
 
-def main(out_dir: str) -> None:
+def main(out_dir: str, device_speeds=None) -> None:
     os.makedirs(out_dir, exist_ok=True)
-    for dtype, speed in (("MI355X", 1.0), ("MI355X_LC", 0.5)):
+    for dtype, speed in device_speeds or (("MI355X", 1.0), ("MI355X_LC", 0.5)):
         for tp in (1, 2, 4):
             for bs in (1, 2, 4):
                 times = layer_times(speed, tp, bs)
