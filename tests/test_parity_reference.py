@@ -224,11 +224,14 @@ def test_het_search_parity_two_device_types(tmp_path):
         parts = line.split(", ")
         if not (parts and parts[0].isdigit()):
             continue
-        # node_sequence itself contains commas: normalize to type names and
-        # take the rest of the row from the device_groups bracket onward
-        seq = tuple(_re.findall(r"DeviceType\.(\w+)", line.split("[", 1)[0]))
+        # node_sequence itself contains commas: compare the row from the
+        # device_groups bracket onward. The reference enumerates
+        # permutations(set(...)) whose ORDER is per-process random (enum
+        # identity hashing), and quirk Q12 ties which sequence carries the
+        # single-stage plans to that order — so the sequence label is not
+        # comparable, while costs and plan content are.
         rest = line[line.index(", [") + 2:]
-        ref_rows.append((round(float(parts[1]), 8), seq, rest))
+        ref_rows.append((round(float(parts[1]), 8), rest))
     assert ref_rows, ref_out[-2000:]
 
     model_from = probe("homo_costs", {
@@ -257,9 +260,7 @@ def test_het_search_parity_two_device_types(tmp_path):
                     min_group_scale_variance=1, max_permute_len=4),
     )
     mine = sorted(
-        (round(r[6], 8), tuple(t.name for t in r[0]),
-         f"{r[1]}, {r[2]}, {r[3]}, {r[4]}")
-        for r in results
+        (round(r[6], 8), f"{r[1]}, {r[2]}, {r[3]}, {r[4]}") for r in results
     )
     theirs = sorted(ref_rows)
     assert len(mine) == len(theirs), (len(mine), len(theirs))
