@@ -74,7 +74,8 @@ def main() -> None:
                 "model": args.model,
             })
         del runner
-        torch.cuda.empty_cache() if torch.cuda.is_available() else None
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
 
     rank = int(os.environ.get("RANK", "0"))
     if rank == 0 and runs:

@@ -124,7 +124,6 @@ def profile_model(
     assert torch.cuda.is_available(), "profiler needs a GPU"
     if seq_length:
         import dataclasses
-        field = "seq_length" if isinstance(spec, LlamaModelSpec) else "seq_length"
         spec = dataclasses.replace(spec, seq_length=seq_length)
     dev = torch.device("cuda", torch.cuda.current_device())
     model_cls = LlamaModel if isinstance(spec, LlamaModelSpec) else GPTModel

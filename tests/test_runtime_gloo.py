@@ -153,3 +153,31 @@ def test_data_parallel_replicas_stay_synced():
 
 def test_pipeline_parallel_step():
     _run_workers(_pp_worker, port=29613)
+
+
+def _plan_runner_cli_worker(rank, world, port, out):
+    _env(rank, world, port)
+    import sys
+
+    from metis_amd.cli import plan_runner
+
+    sys.argv = ["plan_runner", "--model", "gpt2-small", "--plans",
+                "2,1,1,1,2", "--steps", "1", "--warmup", "0",
+                "--out", os.environ["PLAN_OUT"]]
+    # gpt2-small on CPU is heavy; use the tiny spec by patching the registry
+    plan_runner.MODEL_SPECS = dict(plan_runner.MODEL_SPECS)
+    plan_runner.MODEL_SPECS["gpt2-small"] = SPEC
+    plan_runner.main()
+    out.put(("ok", rank))
+
+
+def test_plan_runner_cli_gloo(tmp_path):
+    os.environ["PLAN_OUT"] = str(tmp_path / "measured.json")
+    _run_workers(_plan_runner_cli_worker, port=29614)
+    import json as _json
+
+    doc = _json.loads((tmp_path / "measured.json").read_text())
+    assert len(doc["runs"]) == 1
+    run = doc["runs"][0]
+    assert run["plan"] == {"dp": 2, "tp": 1, "pp": 1, "mbs": 1, "gbs": 2}
+    assert run["measured_ms"] > 0
