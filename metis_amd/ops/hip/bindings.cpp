@@ -43,6 +43,7 @@ torch::Tensor qkv_split_transpose_bwd(
     torch::Tensor dq, torch::Tensor dk, torch::Tensor dv, long head_dim);
 torch::Tensor heads_merge(torch::Tensor x);
 torch::Tensor heads_unmerge(torch::Tensor y, long H);
+torch::Tensor tr16_probe(long stride_elems);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("layernorm_fwd", &layernorm_fwd,
@@ -74,4 +75,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "dq/dk/dv -> fused dqkv layout");
     m.def("heads_merge", &heads_merge, "[B,H,S,D] -> [B,S,H*D]");
     m.def("heads_unmerge", &heads_unmerge, "[B,S,H*D] -> [B,H,S,D]");
+    m.def("tr16_probe", &tr16_probe,
+          "ds_read_b64_tr_b16 lane-semantics probe (debug)");
 }

@@ -247,3 +247,30 @@ torch::Tensor mfma_tile_probe(torch::Tensor a, torch::Tensor b) {
     HIP_CHECK_LAST();
     return d;
 }
+
+// Debug probe: pin ds_read_b64_tr_b16 lane semantics empirically.
+// Fills LDS[i] = i (256 shorts), every lane reads with addr = base +
+// (lane * stride_elems) * 2 bytes, returns the 4 elems per lane.
+namespace {
+typedef short short4v __attribute__((ext_vector_type(4)));
+__global__ void tr16_probe_kernel(short* __restrict__ out, int stride_elems) {
+    __shared__ short lds[256];
+    if (threadIdx.x < 256) lds[threadIdx.x] = (short)threadIdx.x;
+    __syncthreads();
+    const int lane = threadIdx.x & 63;
+    short4v v = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+        (__attribute__((address_space(3))) short4v*)&lds[lane * stride_elems]);
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) out[lane * 4 + j] = v[j];
+}
+}  // namespace
+
+torch::Tensor tr16_probe(long stride_elems) {
+    auto out = torch::empty({64, 4},
+        torch::TensorOptions().dtype(torch::kInt16).device(torch::kCUDA));
+    hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0,
+        c10::hip::getCurrentHIPStream().stream(),
+        reinterpret_cast<short*>(out.data_ptr()), (int)stride_elems);
+    HIP_CHECK_LAST();
+    return out;
+}
