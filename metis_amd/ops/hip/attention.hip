@@ -23,6 +23,8 @@
 
 namespace {
 
+typedef short short4v __attribute__((ext_vector_type(4)));
+
 constexpr int THREADS = 256;
 constexpr int QBLK = 128;     // q rows per block (2 x 16 per wave)
 constexpr int VPAD = 8;       // Vt row padding (bf16 elements)
@@ -68,10 +70,13 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     // raw bf16 BITS as shorts (a short into __hip_bfloat16 converts
-    // numerically — never store element-wise through the struct type)
-    short* Ks = reinterpret_cast<short*>(smem);                 // [64][KSLOT*8]
-    short* Vt = Ks + KVBLK * KSLOT * 8;                         // [D][VROW]
-    short* Pw = Vt + D * VROW + wave * 2 * 16 * VROW;           // [2][16][VROW]
+    // numerically — never store element-wise through the struct type).
+    // V stays in its NATURAL [kv][D] image: PV B-fragments read it with
+    // ds_read_b64_tr_b16 (hardware transpose), so no scalar transpose
+    // staging exists anywhere in this kernel.
+    short* Ks = reinterpret_cast<short*>(smem);                 // [KVBLK][KSLOT*8]
+    short* Vs = Ks + KVBLK * KSLOT * 8;                         // [KVBLK][KSLOT*8]
+    short* Pw = Vs + KVBLK * KSLOT * 8 + wave * 2 * 16 * VROW;  // [2][16][VROW]
 
     // ---- preload Q fragments for both row blocks ------------------------
     bf16x8 q_frag[2][dchunks];
@@ -145,9 +150,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
             *reinterpret_cast<bf16x8*>(Ks + (krow * KSLOT + ks8) * 8) = k_stage[u];
             const int vrow = c % KVBLK;
             const int vs8 = c / KVBLK;
-            #pragma unroll
-            for (int i = 0; i < 8; ++i)
-                Vt[(vs8 * 8 + i) * VROW + vrow] = v_stage[u][i];
+            *reinterpret_cast<bf16x8*>(Vs + (vrow * KSLOT + vs8) * 8) = v_stage[u];
         }
     };
 
@@ -244,14 +247,30 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
         for (int rb = 0; rb < 2; ++rb) {
             if (!rb_active[rb]) continue;
             const short* Prb = Pw + rb * 16 * VROW;
+            // V B-fragments via hardware transpose read: each 16-lane
+            // group cooperatively loads one 4(kv)x16(d) block — lane
+            // l&15 receives its d-column's 4 kv values. Per fragment,
+            // two tr16 reads cover the 8 kv rows of this lane group.
+            const int p4 = lane & 15;     // piece index within the group
             #pragma unroll
             for (int ks = 0; ks < KVBLK / 32; ++ks) {   // 32-wide kv chunks
                 bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(
                     Prb + col16 * VROW + ks * 32 + k8 * 8);
                 #pragma unroll
                 for (int jd = 0; jd < djtiles; ++jd) {
-                    bf16x8 v_frag = *reinterpret_cast<const bf16x8*>(
-                        Vt + (jd * 16 + col16) * VROW + ks * 32 + k8 * 8);
+                    bf16x8 v_frag;
+                    #pragma unroll
+                    for (int r = 0; r < 2; ++r) {
+                        const int kvrow = ks * 32 + k8 * 8 + 4 * r + (p4 >> 2);
+                        const int col = jd * 16 + (p4 & 3) * 4;
+                        short4v t = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                            (__attribute__((address_space(3))) short4v*)(
+                                Vs + kvrow * KSLOT * 8 + col));
+                        v_frag[r * 4 + 0] = t[0];
+                        v_frag[r * 4 + 1] = t[1];
+                        v_frag[r * 4 + 2] = t[2];
+                        v_frag[r * 4 + 3] = t[3];
+                    }
                     o_acc[rb][jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         p_frag, v_frag, o_acc[rb][jd], 0, 0, 0);
                 }
@@ -299,7 +318,7 @@ std::vector<torch::Tensor> attn_fwd(
     #define LAUNCH_D(DD)                                                      \
         do {                                                                  \
             const int kvb = kvblk_for<DD>();                                  \
-            const int lds = (kvb * (DD / 8 + 1) * 8 + DD * (kvb + VPAD)       \
+            const int lds = (2 * kvb * (DD / 8 + 1) * 8                       \
                              + 4 * 2 * 16 * (kvb + VPAD)) * 2;                \
             hipLaunchKernelGGL(attn_fwd_kernel<DD>, dim3(grid),               \
                 dim3(THREADS), lds, stream,                                   \
