@@ -29,11 +29,35 @@
 
 namespace {
 
+typedef short short4v __attribute__((ext_vector_type(4)));
+
 constexpr int THREADS = 256;
 constexpr int RBLK = 128;    // own-side rows per block (2 x 16 per wave)
 constexpr int CTILE = 64;    // opposing-side tile width
 constexpr int VPAD = 8;      // transposed-tile row padding (bf16)
 constexpr int DMAX = 128;
+
+// B-fragment via ds_read_b64_tr_b16 from a NATURAL [row][KSLOT*8] LDS
+// image: per 16-lane group, two reads cover the 8 opposing-index rows;
+// lane l&15 receives its d-column (semantics pinned by ext.tr16_probe).
+template <int KSLOT>
+__device__ __forceinline__ bf16x8 tr16_frag(
+    const short* img, int row0, int col0, int p4) {
+    bf16x8 out;
+    #pragma unroll
+    for (int r = 0; r < 2; ++r) {
+        const int row = row0 + 4 * r + (p4 >> 2);
+        const int col = col0 + (p4 & 3) * 4;
+        short4v t = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (__attribute__((address_space(3))) short4v*)(
+                img + row * KSLOT * 8 + col));
+        out[r * 4 + 0] = t[0];
+        out[r * 4 + 1] = t[1];
+        out[r * 4 + 2] = t[2];
+        out[r * 4 + 3] = t[3];
+    }
+    return out;
+}
 
 template <int D>
 __device__ __forceinline__ bf16x8 frag8(const bf16* base, long row, int d0) {
@@ -77,8 +101,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
     extern __shared__ __attribute__((aligned(16))) char smem[];
     short* Ks = reinterpret_cast<short*>(smem);                    // [CTILE][KSLOT*8]
     short* Vs = Ks + CTILE * KSLOT * 8;                            // [CTILE][KSLOT*8]
-    short* Kt = Vs + CTILE * KSLOT * 8;                            // [D][VROW]
-    short* Sw = Kt + D * VROW + wave * 2 * 16 * VROW;              // [2][16][VROW]
+    short* Sw = Vs + CTILE * KSLOT * 8 + wave * 2 * 16 * VROW;     // [2][16][VROW]
 
     // per-row state and Q/dO fragments for both row blocks
     float lse_r[2][4], delta_r[2][4];
@@ -128,9 +151,6 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
             const int row = c % CTILE;
             const int d0 = (c / CTILE) * 8;
             *reinterpret_cast<bf16x8*>(Ks + row * KSLOT * 8 + d0) = k_stage[u];
-            #pragma unroll
-            for (int i = 0; i < 8; ++i)
-                Kt[(d0 + i) * VROW + row] = k_stage[u][i];
             bf16x8 vv = *reinterpret_cast<const bf16x8*>(
                 V + kv_base + (long)(kv0 + row) * D + d0);
             *reinterpret_cast<bf16x8*>(Vs + row * KSLOT * 8 + d0) = vv;
@@ -202,14 +222,15 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
         for (int rb = 0; rb < 2; ++rb) {
             if (!rb_active[rb]) continue;
             const short* Srb = Sw + rb * 16 * VROW;
+            const int p4 = lane & 15;
             #pragma unroll
             for (int ks = 0; ks < 2; ++ks) {
                 bf16x8 ds_frag = *reinterpret_cast<const bf16x8*>(
                     Srb + col16 * VROW + ks * 32 + k8 * 8);
                 #pragma unroll
                 for (int jd = 0; jd < djtiles; ++jd) {
-                    bf16x8 kt_frag = *reinterpret_cast<const bf16x8*>(
-                        Kt + (jd * 16 + col16) * VROW + ks * 32 + k8 * 8);
+                    bf16x8 kt_frag = tr16_frag<KSLOT>(
+                        Ks, ks * 32 + k8 * 8, jd * 16, p4);
                     dq_acc[rb][jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         ds_frag, kt_frag, dq_acc[rb][jd], 0, 0, 0);
                 }
@@ -264,21 +285,16 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
 
     const int kv0 = kvtile * KVB + wave * 16;   // this wave's kv rows
 
-    // At D <= 96 the natural Q/dO tiles fit in LDS next to the transposed
-    // ones at 2 blocks/CU; at D = 128 that allocation would drop to 1
-    // block/CU, so the natural fragments read straight from global (L2).
-    // Natural Q/dO tiles stay in LDS at every D: the D=128 allocation
-    // drops to 1 block/CU but measures 118 TF/s-eq vs 89 with 2 blocks
-    // reading natural fragments from L2 (A/B on B8 H32 S2048 D128).
-    constexpr bool STAGE_NATURAL = true;
-    constexpr int NATSZ = STAGE_NATURAL ? CTILE * KSLOT * 8 : 0;
+    // Natural Q/dO LDS images only: the dK/dV B-fragments come from
+    // ds_read_b64_tr_b16 straight off these, so the scalar-transposed
+    // Qt/dOt copies (and their LDS staging writes) are gone; the freed
+    // LDS restores 2 blocks/CU at every D.
     extern __shared__ __attribute__((aligned(16))) char smem[];
     short* Qs = reinterpret_cast<short*>(smem);                    // [CTILE][KSLOT*8]
-    short* dOs = Qs + NATSZ;                                       // [CTILE][KSLOT*8]
-    short* Qt = dOs + NATSZ;                                       // [D][VROW]
-    short* dOt = Qt + D * VROW;                                    // [D][VROW]
-    short* Sw = dOt + D * VROW + wave * 16 * VROW;                 // [16][VROW]
-    short* Pw = dOt + D * VROW + 4 * 16 * VROW + wave * 16 * VROW; // [16][VROW]
+    short* dOs = Qs + CTILE * KSLOT * 8;                           // [CTILE][KSLOT*8]
+    short* Sw = dOs + CTILE * KSLOT * 8 + wave * 16 * VROW;        // [16][VROW]
+    short* Pw = dOs + CTILE * KSLOT * 8 + 4 * 16 * VROW
+                + wave * 16 * VROW;                                // [16][VROW]
 
     // K and V fragments for this wave's rows (A layout, m = col16)
     bf16x8 k_frag[dchunks], v_frag[dchunks];
@@ -322,22 +338,14 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
             if (c >= chunks) break;
             const int row = c % CTILE;
             const int d0 = (c / CTILE) * 8;
-            if constexpr (STAGE_NATURAL)
-                *reinterpret_cast<bf16x8*>(Qs + row * KSLOT * 8 + d0) = q_stage[u];
-            #pragma unroll
-            for (int i = 0; i < 8; ++i)
-                Qt[(d0 + i) * VROW + row] = q_stage[u][i];
+            *reinterpret_cast<bf16x8*>(Qs + row * KSLOT * 8 + d0) = q_stage[u];
             bf16x8 dov;
             if constexpr (STAGE_DO)
                 dov = do_stage[u];
             else
                 dov = *reinterpret_cast<const bf16x8*>(
                     dO + q_base + (long)(q0 + row) * D + d0);
-            if constexpr (STAGE_NATURAL)
-                *reinterpret_cast<bf16x8*>(dOs + row * KSLOT * 8 + d0) = dov;
-            #pragma unroll
-            for (int i = 0; i < 8; ++i)
-                dOt[(d0 + i) * VROW + row] = dov[i];
+            *reinterpret_cast<bf16x8*>(dOs + row * KSLOT * 8 + d0) = dov;
         }
     };
 
@@ -361,19 +369,14 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
                 for (int c = 0; c < dchunks; ++c) {
                     const int d0 = c * 32 + k8 * 8;
                     bf16x8 qf, dof;
-                    if constexpr (STAGE_NATURAL) {
-                        if (d0 < D) {
-                            qf = *reinterpret_cast<const bf16x8*>(
-                                Qs + qrow * KSLOT * 8 + d0);
-                            dof = *reinterpret_cast<const bf16x8*>(
-                                dOs + qrow * KSLOT * 8 + d0);
-                        } else {
-                            #pragma unroll
-                            for (int i = 0; i < 8; ++i) { qf[i] = 0; dof[i] = 0; }
-                        }
+                    if (d0 < D) {
+                        qf = *reinterpret_cast<const bf16x8*>(
+                            Qs + qrow * KSLOT * 8 + d0);
+                        dof = *reinterpret_cast<const bf16x8*>(
+                            dOs + qrow * KSLOT * 8 + d0);
                     } else {
-                        qf = frag8<D>(Q + q_base, q0 + qrow, d0);
-                        dof = frag8<D>(dO + q_base, q0 + qrow, d0);
+                        #pragma unroll
+                        for (int i = 0; i < 8; ++i) { qf[i] = 0; dof[i] = 0; }
                     }
                     st_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         k_frag[c], qf, st_acc[j], 0, 0, 0);
@@ -403,6 +406,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
 
             asm volatile("" ::: "memory");   // wave-local publish
 
+            const int p4 = lane & 15;
             #pragma unroll
             for (int ks = 0; ks < 2; ++ks) {
                 bf16x8 dst_frag = *reinterpret_cast<const bf16x8*>(
@@ -411,12 +415,12 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
                     Pw + col16 * VROW + ks * 32 + k8 * 8);
                 #pragma unroll
                 for (int jd = 0; jd < djtiles; ++jd) {
-                    bf16x8 qt_frag = *reinterpret_cast<const bf16x8*>(
-                        Qt + (jd * 16 + col16) * VROW + ks * 32 + k8 * 8);
+                    bf16x8 qt_frag = tr16_frag<KSLOT>(
+                        Qs, ks * 32 + k8 * 8, jd * 16, p4);
                     dk_acc[jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         dst_frag, qt_frag, dk_acc[jd], 0, 0, 0);
-                    bf16x8 dot_frag = *reinterpret_cast<const bf16x8*>(
-                        dOt + (jd * 16 + col16) * VROW + ks * 32 + k8 * 8);
+                    bf16x8 dot_frag = tr16_frag<KSLOT>(
+                        dOs, ks * 32 + k8 * 8, jd * 16, p4);
                     dv_acc[jd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         pt_frag, dot_frag, dv_acc[jd], 0, 0, 0);
                 }
@@ -465,9 +469,8 @@ std::vector<torch::Tensor> attn_bwd(
             const int vrow = CTILE + VPAD;                                    \
             const int lds_dq = (2 * CTILE * (DD / 8 + 1) * 8                  \
                                 + DD * vrow + 4 * 2 * 16 * vrow) * 2;         \
-            const int nat = 2 * CTILE * (DD / 8 + 1) * 8;                     \
-            const int lds_dkv = (nat                                          \
-                                 + 2 * DD * vrow + 2 * 4 * 16 * vrow) * 2;    \
+            const int lds_dkv = (2 * CTILE * (DD / 8 + 1) * 8                 \
+                                 + 2 * 4 * 16 * vrow) * 2;                    \
             hipLaunchKernelGGL(attn_bwd_dq_kernel<DD>, dim3(grid),            \
                 dim3(THREADS), lds_dq, stream,                                \
                 reinterpret_cast<const bf16*>(qc.data_ptr()),                 \
