@@ -22,3 +22,13 @@ def bench(fn, iters=20):
 fl = 4 * B * H * S * S * D / 2 * 2.5
 t = bench(lambda: ext.attn_bwd(q, k, v, do, lse, delta, sc))
 print(f"attn bwd D80: {fl/t/1e12:.0f} TF/s-equiv")
+
+B, H, S, D = 4, 32, 2048, 128
+q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+k = torch.randn_like(q); v = torch.randn_like(q); do = torch.randn_like(q)
+sc = 1 / math.sqrt(D)
+o, lse = ext.attn_fwd(q, k, v, sc)
+delta = (do.float() * o.float()).sum(-1).contiguous()
+fl = 4 * B * H * S * S * D / 2 * 2.5
+t = bench(lambda: ext.attn_bwd(q, k, v, do, lse, delta, sc))
+print(f"attn bwd D128: {fl/t/1e12:.0f} TF/s-equiv")
