@@ -47,6 +47,8 @@ def parse_args():
                         "--profile-dir instead of the flags")
     p.add_argument("--profile-dir", default=None,
                    help="defaults to profiles/mi355x/<model>")
+    p.add_argument("--schedule", default="gpipe", choices=("gpipe", "1f1b"))
+    p.add_argument("--recompute", action="store_true")
     return p.parse_args()
 
 
@@ -87,7 +89,8 @@ def main() -> None:
                                    comm_bench_path="profiles/comm_bench.json")
 
     ctx = init_parallel(dp=dp, tp=tp, pp=pp)
-    runner = PlanRunner(spec, ctx, mbs=mbs, gbs=gbs)
+    runner = PlanRunner(spec, ctx, mbs=mbs, gbs=gbs,
+                        schedule=args.schedule, recompute=args.recompute)
 
     ms = runner.timed_steps(args.steps, args.warmup)
 

@@ -49,6 +49,9 @@ def main() -> None:
     p.add_argument("--steps", type=int, default=5)
     p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--out", default="profiles/measured_runs.json")
+    p.add_argument("--schedule", default="gpipe", choices=("gpipe", "1f1b"))
+    p.add_argument("--recompute", action="store_true",
+                   help="per-block activation recomputation")
     args = p.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -59,7 +62,9 @@ def main() -> None:
             print(f"skip dp{dp}_tp{tp}_pp{pp}: needs world {dp * tp * pp}, have {world}")
             continue
         ctx = init_parallel(dp=dp, tp=tp, pp=pp)
-        runner = PlanRunner(spec, ctx, mbs=mbs, gbs=gbs)
+        runner = PlanRunner(spec, ctx, mbs=mbs, gbs=gbs,
+                            schedule=args.schedule,
+                            recompute=args.recompute)
         ms = runner.timed_steps(args.steps, args.warmup)
         if dist.is_initialized():
             t = torch.tensor([ms], dtype=torch.float64,

@@ -18,6 +18,7 @@ from dataclasses import dataclass
 from typing import List, Optional
 
 import torch
+import torch.utils.checkpoint
 import torch.distributed as dist
 import torch.nn as nn
 import torch.nn.functional as F
@@ -256,6 +257,11 @@ class GPTModel(nn.Module):
             self.ln_final = LayerNorm(h)
             self.head = ColumnParallelLinear(h, spec.vocab_size, tp, dtype)
 
+        # activation recomputation: re-run each block's forward during
+        # backward instead of keeping its activations (trades ~30% extra
+        # compute for O(1) per-block activation memory)
+        self.recompute = False
+
     def forward(
         self, x: torch.Tensor, labels: Optional[torch.Tensor] = None
     ) -> torch.Tensor:
@@ -267,8 +273,13 @@ class GPTModel(nn.Module):
             pos = torch.arange(s, device=x.device)
             x = self.wte(x) + self.wpe(pos)[None, :, :]
 
+        use_ckpt = self.recompute and torch.is_grad_enabled()
         for block in self.blocks:
-            x = block(x, self.tp_group)
+            if use_ckpt:
+                x = torch.utils.checkpoint.checkpoint(
+                    block, x, self.tp_group, use_reentrant=False)
+            else:
+                x = block(x, self.tp_group)
 
         if self.has_head:
             x = self.ln_final(x)

@@ -9,6 +9,7 @@ from dataclasses import dataclass
 from typing import List, Optional
 
 import torch
+import torch.utils.checkpoint
 import torch.distributed as dist
 import torch.nn as nn
 
@@ -139,11 +140,18 @@ class LlamaModel(nn.Module):
             self.norm_final = RMSNorm(h)
             self.head = ColumnParallelLinear(h, spec.vocab_size, tp, dtype)
 
+        self.recompute = False  # see GPTModel.recompute
+
     def forward(self, x, labels=None):
         if self.has_embedding:
             x = self.wte(x)
+        use_ckpt = self.recompute and torch.is_grad_enabled()
         for block in self.blocks:
-            x = block(x, self.tp_group)
+            if use_ckpt:
+                x = torch.utils.checkpoint.checkpoint(
+                    block, x, self.tp_group, use_reentrant=False)
+            else:
+                x = block(x, self.tp_group)
         if self.has_head:
             x = self.norm_final(x)
             logits = self.head(x, self.tp_group)

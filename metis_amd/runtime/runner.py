@@ -38,7 +38,11 @@ class PlanRunner:
         layer_partition: Optional[List[int]] = None,
         lr: float = 1e-4,
         dtype: torch.dtype = torch.bfloat16,
+        schedule: str = "gpipe",
+        recompute: bool = False,
     ) -> None:
+        assert schedule in ("gpipe", "1f1b")
+        self.schedule = schedule
         self.spec = spec
         self.ctx = ctx
         self.mbs = mbs
@@ -63,6 +67,7 @@ class PlanRunner:
             spec, tp=ctx.tp, dtype=dtype, layer_range=(start, end),
             tp_group=ctx.tp_group,
         )
+        self.model.recompute = recompute
         if ctx.device is not None:
             self.model.to(ctx.device)
         self.optimizer = FusedAdamW(self.model.parameters(), lr=lr)
@@ -171,6 +176,70 @@ class PlanRunner:
             return float(torch.stack([l.detach() for l in losses]).mean())
         return 0.0
 
+    def _step_pipeline_1f1b(self) -> float:
+        """1F1B (one-forward-one-backward): same bubble as GPipe
+        ((B-1)*max + sum) but at most pp - pp_rank activations are live per
+        stage instead of all B — the memory-bound pipeline schedule the
+        reference's cost model prices but cannot run. Sends are isend (a
+        steady-state fwd-send/bwd-send pair of neighboring stages would
+        deadlock with rendezvous sends); recvs stay blocking."""
+        ctx = self.ctx
+        nm = self.num_microbatches
+        act_shape = (self.mbs, self.spec.seq_length, self.spec.hidden_size)
+        prev = ctx.stage_neighbor(-1) if not ctx.is_first_stage else None
+        nxt = ctx.stage_neighbor(+1) if not ctx.is_last_stage else None
+        self.optimizer.zero_grad()
+
+        pending = []   # (work, tensor): keep the buffer alive until wait
+        live = {}      # microbatch -> (stage input, stage output)
+        losses: List[torch.Tensor] = []
+
+        def fwd(i: int) -> None:
+            inp = None
+            if ctx.is_first_stage:
+                x, _ = self.synthetic_batch()
+            else:
+                x = self._recv_activation(act_shape, prev).requires_grad_(True)
+                inp = x
+            if ctx.is_last_stage:
+                _, labels = self.synthetic_batch()
+                out = self.model(x, labels=labels)
+                losses.append(out.detach())
+            else:
+                out = self.model(x)
+                t = out.detach().contiguous()
+                pending.append((dist.isend(t, dst=nxt), t))
+            live[i] = (inp, out)
+
+        def bwd(i: int) -> None:
+            if self.grad_sync is not None and i == nm - 1:
+                self.grad_sync.arm()
+            inp, out = live.pop(i)
+            if ctx.is_last_stage:
+                (out / nm).backward()
+            else:
+                gout = self._recv_activation(act_shape, nxt)
+                out.backward(gout)
+            if not ctx.is_first_stage:
+                t = inp.grad.contiguous()
+                pending.append((dist.isend(t, dst=prev), t))
+
+        warm = min(nm, ctx.pp - 1 - ctx.pp_rank)
+        for i in range(warm):
+            fwd(i)
+        for i in range(warm, nm):           # steady state
+            fwd(i)
+            bwd(i - warm)
+        for i in range(nm - warm, nm):      # drain
+            bwd(i)
+        for work, _ in pending:
+            work.wait()
+
+        self._sync_and_step()
+        if losses:
+            return float(torch.stack(losses).mean())
+        return 0.0
+
     # --- gradient sync + optimizer ----------------------------------------
     def _sync_and_step(self) -> None:
         with self.tracer.span("grad_sync"):
@@ -186,6 +255,8 @@ class PlanRunner:
     def train_step(self) -> float:
         if self.ctx.pp == 1:
             return self._step_no_pipeline()
+        if self.schedule == "1f1b":
+            return self._step_pipeline_1f1b()
         return self._step_pipeline()
 
     # --- checkpoint ---------------------------------------------------------
