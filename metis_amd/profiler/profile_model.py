@@ -19,6 +19,7 @@ import time
 from typing import Dict, List, Optional
 
 import torch
+import torch.utils.checkpoint
 import torch.distributed as dist
 
 from metis_amd.models.gpt import GPTModel, GPTModelSpec, MODEL_SPECS as _GPT_SPECS
@@ -119,8 +120,17 @@ def profile_model(
     iters: int = 10,
     tp_group=None,
     seq_length: Optional[int] = None,
+    recompute: bool = False,
 ) -> Optional[str]:
-    """Profile one (tp, bs) point; returns the JSON path (rank 0)."""
+    """Profile one (tp, bs) point; returns the JSON path (rank 0).
+
+    With ``recompute=True`` the blocks run under activation
+    recomputation, so the memory (and time) numbers reflect that
+    execution mode — plan search against such a profile directory is
+    how the planner prices recompute (measured, not modeled). Per-block
+    times are then a uniform split of the block total: module backward
+    hooks double-fire during checkpoint replay, and the blocks are
+    identical anyway."""
     assert torch.cuda.is_available(), "profiler needs a GPU"
     if seq_length:
         import dataclasses
@@ -152,7 +162,11 @@ def profile_model(
         t0 = time.perf_counter()
         x = embed(tokens)
         for blk in model.blocks:
-            x = blk(x, model.tp_group)
+            if recompute:
+                x = torch.utils.checkpoint.checkpoint(
+                    blk, x, model.tp_group, use_reentrant=False)
+            else:
+                x = blk(x, model.tp_group)
         loss = head(x, labels)
         loss.backward()
         torch.cuda.synchronize()
@@ -209,6 +223,12 @@ def profile_model(
     for k in acc:
         acc[k] /= iters
     layer_ms = [v / iters for v in layer_acc]
+    if recompute and len(layer_ms) > 2:
+        # block event brackets are unreliable under checkpoint replay;
+        # split the block total uniformly (identical blocks)
+        block_total = acc["fwd_bwd"] - layer_ms[0] - layer_ms[-1]
+        nb = len(layer_ms) - 2
+        layer_ms = [layer_ms[0]] + [max(block_total, 0.0) / nb] * nb + [layer_ms[-1]]
 
     # per-layer memory: parameter + grad + optimizer state bytes + peak
     # activation delta measured by the forward hooks
@@ -257,6 +277,9 @@ def main() -> None:
     p.add_argument("--iters", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--seq-length", type=int, default=None)
+    p.add_argument("--recompute", action="store_true",
+                   help="profile with per-block activation recomputation "
+                        "(default out dir: profiles/mi355x_rc/<model>)")
     args = p.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -268,12 +291,14 @@ def main() -> None:
         tp_group = ctx.tp_group
 
     spec = MODEL_SPECS[args.model]
-    out_dir = args.out or f"profiles/mi355x/{args.model}"
+    base = "profiles/mi355x_rc" if args.recompute else "profiles/mi355x"
+    out_dir = args.out or f"{base}/{args.model}"
     for bs in [int(b) for b in args.bs.split(",")]:
         path = profile_model(
             spec, bs=bs, tp=world, device_type=args.device_type,
             out_dir=out_dir, warmup=args.warmup, iters=args.iters,
             tp_group=tp_group, seq_length=args.seq_length,
+            recompute=args.recompute,
         )
         if path:
             print(f"wrote {path}")
