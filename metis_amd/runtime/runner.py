@@ -13,6 +13,7 @@ AdamW kernel then consumes the flat fp32 buffer directly.
 
 from __future__ import annotations
 
+import os
 import time
 from typing import List, Optional, Tuple
 
@@ -73,6 +74,7 @@ class PlanRunner:
         self.optimizer = FusedAdamW(self.model.parameters(), lr=lr)
         self.dtype = dtype
         self.tracer = tracer_from_env(ctx.rank)
+        self._check_sync = os.environ.get("METIS_CHECK_SYNC") == "1"
         self._data_gen = None
         # bucketed, overlapped DP gradient all-reduce (dp > 1 only: the
         # per-parameter Python hooks cost more than the serial gather saves
@@ -251,6 +253,25 @@ class PlanRunner:
                 self.optimizer.gather_grads()
         with self.tracer.span("optimizer"):
             self.optimizer.step(pre_gathered=True)
+        if self._check_sync:
+            self.verify_replicas_synced()
+
+    def verify_replicas_synced(self) -> None:
+        """Race detector for the DP path (METIS_CHECK_SYNC=1): bitwise
+        checksum of the fp32 master weights compared across the DP group
+        after the optimizer step; raises on divergence. The reference has
+        no runtime at all, so no equivalent exists there."""
+        if self.ctx.dp <= 1 or self.ctx.dp_group is None:
+            return
+        digest = self.optimizer.master.view(torch.int32).long().sum()[None]
+        gathered = [torch.zeros_like(digest) for _ in range(self.ctx.dp)]
+        dist.all_gather(gathered, digest, group=self.ctx.dp_group)
+        if any(int(g[0]) != int(gathered[0][0]) for g in gathered):
+            raise RuntimeError(
+                "DP replicas diverged after optimizer step: master "
+                f"checksums {[int(g[0]) for g in gathered]} "
+                f"(dp_rank {self.ctx.dp_rank})"
+            )
 
     def train_step(self) -> float:
         if self.ctx.pp == 1:

@@ -208,3 +208,40 @@ def _pp_1f1b_worker(rank, world, port, out):
 
 def test_pipeline_1f1b_matches_gpipe():
     _run_workers(_pp_1f1b_worker, port=29615)
+
+
+def _dp_sync_check_worker(rank, world, port, out):
+    _env(rank, world, port)
+    os.environ["METIS_CHECK_SYNC"] = "1"
+    from metis_amd.runtime.comm import init_parallel
+    from metis_amd.runtime.runner import PlanRunner
+
+    ctx = init_parallel(dp=2, tp=1, pp=1)
+    torch.manual_seed(3)
+    runner = PlanRunner(SPEC, ctx, mbs=2, gbs=8, dtype=torch.float32)
+    torch.manual_seed(100 + rank)
+    runner.train_step()  # checker passes on a healthy step
+
+    # inject a divergence on rank 1: the checker must raise on ALL ranks
+    if rank == 1:
+        with torch.no_grad():
+            runner.optimizer.master[0] += 1.0
+    try:
+        runner.verify_replicas_synced()
+        out.put(("no-raise", rank))
+    except RuntimeError:
+        out.put(("raised", rank))
+    dist.destroy_process_group()
+
+
+def test_dp_sync_checker_detects_divergence():
+    mp_ctx = mp.get_context("spawn")
+    out = mp_ctx.Queue()
+    procs = [mp_ctx.Process(target=_dp_sync_check_worker,
+                            args=(r, 2, 29616, out)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+    results = [out.get() for _ in range(2)]
+    assert all(r[0] == "raised" for r in results), results
