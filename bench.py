@@ -25,8 +25,9 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 from metis_amd.models.gpt import MODEL_SPECS as _GPT_SPECS  # noqa: E402
 from metis_amd.models.llama import LLAMA_SPECS  # noqa: E402
+from metis_amd.models.moe import MOE_SPECS  # noqa: E402
 
-MODEL_SPECS = {**_GPT_SPECS, **LLAMA_SPECS}
+MODEL_SPECS = {**_GPT_SPECS, **LLAMA_SPECS, **MOE_SPECS}
 from metis_amd.runtime.comm import init_parallel  # noqa: E402
 from metis_amd.runtime.runner import PlanRunner  # noqa: E402
 

@@ -21,9 +21,10 @@ import torch
 import torch.distributed as dist
 
 from metis_amd.models.gpt import MODEL_SPECS as _GPT_SPECS
-from metis_amd.models.llama import LLAMA_SPECS
+from metis_amd.models.llama import LLAMA_SPECS  # noqa: E402
+from metis_amd.models.moe import MOE_SPECS  # noqa: E402
 
-MODEL_SPECS = {**_GPT_SPECS, **LLAMA_SPECS}
+MODEL_SPECS = {**_GPT_SPECS, **LLAMA_SPECS, **MOE_SPECS}
 from metis_amd.planner.validate import plan_key
 from metis_amd.runtime.comm import init_parallel
 from metis_amd.runtime.runner import PlanRunner

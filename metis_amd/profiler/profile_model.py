@@ -24,8 +24,9 @@ import torch.distributed as dist
 
 from metis_amd.models.gpt import GPTModel, GPTModelSpec, MODEL_SPECS as _GPT_SPECS
 from metis_amd.models.llama import LlamaModel, LlamaModelSpec, LLAMA_SPECS
+from metis_amd.models.moe import MoEModel, MoEModelSpec, MOE_SPECS
 
-MODEL_SPECS = {**_GPT_SPECS, **LLAMA_SPECS}
+MODEL_SPECS = {**_GPT_SPECS, **LLAMA_SPECS, **MOE_SPECS}
 from metis_amd.ops import FusedAdamW
 from metis_amd.profiles import ProfileStore
 
@@ -136,7 +137,12 @@ def profile_model(
         import dataclasses
         spec = dataclasses.replace(spec, seq_length=seq_length)
     dev = torch.device("cuda", torch.cuda.current_device())
-    model_cls = LlamaModel if isinstance(spec, LlamaModelSpec) else GPTModel
+    if isinstance(spec, MoEModelSpec):
+        model_cls = MoEModel
+    elif isinstance(spec, LlamaModelSpec):
+        model_cls = LlamaModel
+    else:
+        model_cls = GPTModel
     model = model_cls(spec, tp=tp, dtype=torch.bfloat16, tp_group=tp_group).to(dev)
     opt = FusedAdamW(model.parameters(), lr=1e-4)
 

@@ -22,6 +22,7 @@ import torch.distributed as dist
 
 from metis_amd.models.gpt import GPTModel, GPTModelSpec
 from metis_amd.models.llama import LlamaModel, LlamaModelSpec
+from metis_amd.models.moe import MoEModel, MoEModelSpec
 from metis_amd.ops import FusedAdamW
 from metis_amd.planner.volume import uniform_layer_split
 from metis_amd.runtime.comm import ParallelContext
@@ -63,7 +64,12 @@ class PlanRunner:
         self.layer_partition = layer_partition
         start, end = layer_partition[ctx.pp_rank], layer_partition[ctx.pp_rank + 1]
 
-        model_cls = LlamaModel if isinstance(spec, LlamaModelSpec) else GPTModel
+        if isinstance(spec, MoEModelSpec):
+            model_cls = MoEModel
+        elif isinstance(spec, LlamaModelSpec):
+            model_cls = LlamaModel
+        else:
+            model_cls = GPTModel
         self.model = model_cls(
             spec, tp=ctx.tp, dtype=dtype, layer_range=(start, end),
             tp_group=ctx.tp_group,
@@ -124,6 +130,22 @@ class PlanRunner:
         dist.recv(buf, src=src)
         return buf
 
+    def _stage_aux(self) -> Optional[torch.Tensor]:
+        """Routing aux loss of the microbatch just forwarded (MoE stages
+        without the head; stages with the head fold it into the loss)."""
+        consume = getattr(self.model, "consume_aux_loss", None)
+        return consume() if consume is not None else None
+
+    def _backward_stage(self, out, gout, aux) -> None:
+        """Non-last-stage backward incl. this stage's aux loss in the same
+        engine pass (the logical loss is mean_mb(ce + aux))."""
+        if aux is not None:
+            torch.autograd.backward(
+                [out, aux],
+                [gout, torch.full_like(aux, 1.0 / self.num_microbatches)])
+        else:
+            out.backward(gout)
+
     def _step_pipeline(self) -> float:
         ctx = self.ctx
         h = self.spec.hidden_size
@@ -135,7 +157,7 @@ class PlanRunner:
         inputs: List[Optional[torch.Tensor]] = []
         outputs: List[torch.Tensor] = []
         losses: List[torch.Tensor] = []
-        label_cache: List[torch.Tensor] = []
+        aux_terms: List[Optional[torch.Tensor]] = []
 
         # forward fill
         for _ in range(self.num_microbatches):
@@ -159,6 +181,7 @@ class PlanRunner:
             else:
                 out = self.model(x)
                 outputs.append(out)
+                aux_terms.append(self._stage_aux())
                 dist.send(out.detach().contiguous(), dst=nxt)
 
         # backward drain (reverse order)
@@ -169,7 +192,7 @@ class PlanRunner:
                 (outputs[i] / self.num_microbatches).backward()
             else:
                 gout = self._recv_activation(act_shape, nxt)
-                outputs[i].backward(gout)
+                self._backward_stage(outputs[i], gout, aux_terms[i])
             if not ctx.is_first_stage:
                 dist.send(inputs[i].grad.contiguous(), dst=prev)
 
@@ -211,17 +234,17 @@ class PlanRunner:
                 out = self.model(x)
                 t = out.detach().contiguous()
                 pending.append((dist.isend(t, dst=nxt), t))
-            live[i] = (inp, out)
+            live[i] = (inp, out, self._stage_aux())
 
         def bwd(i: int) -> None:
             if self.grad_sync is not None and i == nm - 1:
                 self.grad_sync.arm()
-            inp, out = live.pop(i)
+            inp, out, aux = live.pop(i)
             if ctx.is_last_stage:
                 (out / nm).backward()
             else:
                 gout = self._recv_activation(act_shape, nxt)
-                out.backward(gout)
+                self._backward_stage(out, gout, aux)
             if not ctx.is_first_stage:
                 t = inp.grad.contiguous()
                 pending.append((dist.isend(t, dst=prev), t))

@@ -1,0 +1,70 @@
+"""Single-process MoE checks: routing math, aux loss, recompute parity,
+and a full PlanRunner step (dp=tp=pp=1)."""
+
+import torch
+
+from metis_amd.models.moe import MoEModel, MOE_SPECS
+
+SPEC = MOE_SPECS["moe-tiny"]
+
+
+def _batch():
+    g = torch.Generator().manual_seed(11)
+    tokens = torch.randint(0, SPEC.vocab_size, (2, 32), generator=g)
+    return tokens, torch.roll(tokens, -1, 1)
+
+
+def test_moe_forward_backward():
+    torch.manual_seed(0)
+    m = MoEModel(SPEC, dtype=torch.float32)
+    tokens, labels = _batch()
+    loss = m(tokens, labels=labels)
+    assert float(loss) > 0
+    loss.backward()
+    for n, p in m.named_parameters():
+        assert p.grad is not None, n
+    # every expert of every block saw tokens -> non-zero grads (top-2 of 4
+    # experts over 64 tokens makes an unused expert vanishingly unlikely)
+    for b in m.blocks:
+        per_expert = b.experts.w1.grad.flatten(1).abs().sum(1)
+        assert (per_expert > 0).all()
+
+
+def test_moe_aux_loss_range():
+    torch.manual_seed(0)
+    m = MoEModel(SPEC, dtype=torch.float32)
+    tokens, labels = _batch()
+    m(tokens)  # no labels: aux left on the blocks
+    aux = m.consume_aux_loss()
+    # Switch aux is >= 1 (perfect balance) and <= num_experts; coef applied
+    assert SPEC.aux_loss_coef * 0.99 <= float(aux) <= SPEC.aux_loss_coef * SPEC.num_experts
+    assert m.consume_aux_loss() is None  # consumed
+
+
+def test_moe_recompute_grad_parity():
+    torch.manual_seed(0)
+    m = MoEModel(SPEC, dtype=torch.float32)
+    tokens, labels = _batch()
+    loss = m(tokens, labels=labels)
+    loss.backward()
+    ref = {n: p.grad.clone() for n, p in m.named_parameters()}
+    for p in m.parameters():
+        p.grad = None
+    m.recompute = True
+    loss2 = m(tokens, labels=labels)
+    loss2.backward()
+    assert torch.equal(loss.detach(), loss2.detach())
+    for n, p in m.named_parameters():
+        assert torch.allclose(ref[n], p.grad, atol=1e-6), n
+
+
+def test_moe_plan_runner_step():
+    from metis_amd.runtime.comm import ParallelContext
+    from metis_amd.runtime.runner import PlanRunner
+
+    ctx = ParallelContext(rank=0, world_size=1, local_rank=0, dp=1, tp=1, pp=1)
+    torch.manual_seed(3)
+    runner = PlanRunner(SPEC, ctx, mbs=2, gbs=4, dtype=torch.float32)
+    l1 = runner.train_step()
+    l2 = runner.train_step()
+    assert l1 > 0 and l2 > 0
