@@ -50,6 +50,8 @@ def parse_args():
                    help="defaults to profiles/mi355x/<model>")
     p.add_argument("--schedule", default="gpipe", choices=("gpipe", "1f1b"))
     p.add_argument("--recompute", action="store_true")
+    p.add_argument("--zero1", action="store_true",
+                   help="shard optimizer state over the DP group")
     return p.parse_args()
 
 
@@ -91,7 +93,8 @@ def main() -> None:
 
     ctx = init_parallel(dp=dp, tp=tp, pp=pp)
     runner = PlanRunner(spec, ctx, mbs=mbs, gbs=gbs,
-                        schedule=args.schedule, recompute=args.recompute)
+                        schedule=args.schedule, recompute=args.recompute,
+                        zero1=args.zero1)
 
     ms = runner.timed_steps(args.steps, args.warmup)
 

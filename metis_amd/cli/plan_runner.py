@@ -53,6 +53,8 @@ def main() -> None:
     p.add_argument("--schedule", default="gpipe", choices=("gpipe", "1f1b"))
     p.add_argument("--recompute", action="store_true",
                    help="per-block activation recomputation")
+    p.add_argument("--zero1", action="store_true",
+                   help="shard optimizer state over the DP group")
     args = p.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -65,7 +67,8 @@ def main() -> None:
         ctx = init_parallel(dp=dp, tp=tp, pp=pp)
         runner = PlanRunner(spec, ctx, mbs=mbs, gbs=gbs,
                             schedule=args.schedule,
-                            recompute=args.recompute)
+                            recompute=args.recompute,
+                            zero1=args.zero1)
         ms = runner.timed_steps(args.steps, args.warmup)
         if dist.is_initialized():
             t = torch.tensor([ms], dtype=torch.float64,

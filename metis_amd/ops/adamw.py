@@ -3,13 +3,21 @@
 One flat fp32 master buffer + flat m/v moments per parameter group; the
 fused gfx950 kernel updates master, moments and the bf16 working copy in
 one pass. On CPU the same math runs in eager PyTorch (tests, gloo runs).
+
+ZeRO-1 (``shard_group``): master/m/v are sharded 1/ws per rank of the
+group (the DP group in practice — 12 B/param of optimizer state becomes
+12/ws); each rank updates its shard and an all-gather of the updated
+weights replaces the post-step scatter. Gradients still arrive via the
+ordinary bucketed DP all-reduce (same wire bytes as reduce-scatter +
+the weight all-gather on a ring).
 """
 
 from __future__ import annotations
 
-from typing import Iterable, List
+from typing import Iterable, List, Optional
 
 import torch
+import torch.distributed as dist
 
 from metis_amd import ops as _ops
 
@@ -22,6 +30,7 @@ class FusedAdamW:
         betas=(0.9, 0.95),
         eps: float = 1e-8,
         weight_decay: float = 0.1,
+        shard_group: Optional[object] = None,
     ) -> None:
         self.params: List[torch.nn.Parameter] = [p for p in params if p.requires_grad]
         if not self.params:
@@ -31,23 +40,33 @@ class FusedAdamW:
         self.eps = eps
         self.weight_decay = weight_decay
         self.step_count = 0
+        self.shard_group = shard_group
+        ws = dist.get_world_size(shard_group) if shard_group is not None else 1
+        self._shard_ws = ws
+        self._shard_rank = dist.get_rank(shard_group) if shard_group is not None else 0
 
         dev = self.params[0].device
-        # pad the flat buffers to a multiple of 4 for the 16 B kernel path
+        # pad the flat buffers to a multiple of 4 (16 B kernel path) x ws
         total = sum(p.numel() for p in self.params)
-        self._n = (total + 3) // 4 * 4
-        self.master = torch.zeros(self._n, dtype=torch.float32, device=dev)
-        self.m = torch.zeros_like(self.master)
-        self.v = torch.zeros_like(self.master)
-        self._grad_flat = torch.zeros_like(self.master)
+        self._n = (total + 4 * ws - 1) // (4 * ws) * (4 * ws)
+        self._shard_n = self._n // ws
+        self._shard_off = self._shard_rank * self._shard_n
+        self._grad_flat = torch.zeros(self._n, dtype=torch.float32, device=dev)
 
+        full = torch.zeros(self._n, dtype=torch.float32, device=dev)
         offset = 0
         self._slices = []
         for p in self.params:
             n = p.numel()
-            self.master[offset:offset + n].copy_(p.detach().reshape(-1).float())
+            full[offset:offset + n].copy_(p.detach().reshape(-1).float())
             self._slices.append((offset, n))
             offset += n
+        if ws > 1:
+            self.master = full[self._shard_off:self._shard_off + self._shard_n].clone()
+        else:
+            self.master = full
+        self.m = torch.zeros_like(self.master)
+        self.v = torch.zeros_like(self.master)
 
         # On GPU with all-bf16 params, re-bind every parameter to a view of
         # one flat bf16 buffer: the fused kernel then writes the bf16 copy
@@ -87,23 +106,16 @@ class FusedAdamW:
     def gather_grads(self) -> torch.Tensor:
         return self._gather_grads()
 
-    @torch.no_grad()
-    def step(self, grad_scale: float = 1.0, pre_gathered: bool = False) -> None:
-        """Apply one AdamW step. With ``pre_gathered=True`` the caller has
-        already filled (and possibly all-reduced) ``grad_flat``."""
-        self.step_count += 1
-        grads = self._grad_flat if pre_gathered else self._gather_grads()
-
+    def _update(self, grads: torch.Tensor, model: torch.Tensor,
+                grad_scale: float) -> None:
+        """AdamW update of (master, m, v) from ``grads`` (same length)."""
         if self.master.is_cuda:
             ext = _ops.require_extension()
-            model = self._model_flat if self._model_flat is not None else torch.Tensor()
             ext.adamw_step(
                 self.master, model, grads, self.m, self.v,
                 self.lr, self.beta1, self.beta2, self.eps,
                 self.weight_decay, self.step_count, grad_scale,
             )
-            if self._model_flat is not None:
-                return
         else:
             g = grads * grad_scale
             self.m.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
@@ -115,6 +127,45 @@ class FusedAdamW:
                             + self.weight_decay * self.master)
             )
 
+    def _gather_weight_shards(self) -> torch.Tensor:
+        """All-gather the updated fp32 master shards into one full buffer."""
+        full = torch.empty(self._n, dtype=torch.float32,
+                           device=self.master.device)
+        try:
+            dist.all_gather_into_tensor(full, self.master, group=self.shard_group)
+        except (RuntimeError, ValueError):
+            parts = [torch.empty_like(self.master) for _ in range(self._shard_ws)]
+            dist.all_gather(parts, self.master, group=self.shard_group)
+            full = torch.cat(parts)
+        return full
+
+    @torch.no_grad()
+    def step(self, grad_scale: float = 1.0, pre_gathered: bool = False) -> None:
+        """Apply one AdamW step. With ``pre_gathered=True`` the caller has
+        already filled (and possibly all-reduced) ``grad_flat``."""
+        self.step_count += 1
+        grads = self._grad_flat if pre_gathered else self._gather_grads()
+
+        if self._shard_ws > 1:  # ZeRO-1: update the local shard, gather weights
+            off, n = self._shard_off, self._shard_n
+            model = (self._model_flat[off:off + n]
+                     if self._model_flat is not None else torch.Tensor())
+            self._update(grads[off:off + n], model, grad_scale)
+            if self._model_flat is not None:
+                # bf16 weight all-gather straight into the flat param buffer
+                shard = self._model_flat[off:off + n].clone()
+                dist.all_gather_into_tensor(self._model_flat, shard,
+                                            group=self.shard_group)
+                return
+            full = self._gather_weight_shards()
+            for p, (o, pn) in zip(self.params, self._slices):
+                p.copy_(full[o:o + pn].view_as(p).to(p.dtype))
+            return
+
+        model = self._model_flat if self._model_flat is not None else torch.Tensor()
+        self._update(grads, model, grad_scale)
+        if self._model_flat is not None:
+            return
         # scatter master back into the (possibly bf16) working params
         for p, (off, n) in zip(self.params, self._slices):
             p.copy_(self.master[off:off + n].view_as(p).to(p.dtype))

@@ -42,6 +42,7 @@ class PlanRunner:
         dtype: torch.dtype = torch.bfloat16,
         schedule: str = "gpipe",
         recompute: bool = False,
+        zero1: bool = False,
     ) -> None:
         assert schedule in ("gpipe", "1f1b")
         self.schedule = schedule
@@ -77,7 +78,9 @@ class PlanRunner:
         self.model.recompute = recompute
         if ctx.device is not None:
             self.model.to(ctx.device)
-        self.optimizer = FusedAdamW(self.model.parameters(), lr=lr)
+        shard_group = ctx.dp_group if (zero1 and ctx.dp > 1) else None
+        self.optimizer = FusedAdamW(self.model.parameters(), lr=lr,
+                                    shard_group=shard_group)
         self.dtype = dtype
         self.tracer = tracer_from_env(ctx.rank)
         self._check_sync = os.environ.get("METIS_CHECK_SYNC") == "1"
@@ -286,12 +289,17 @@ class PlanRunner:
         no runtime at all, so no equivalent exists there."""
         if self.ctx.dp <= 1 or self.ctx.dp_group is None:
             return
-        digest = self.optimizer.master.view(torch.int32).long().sum()[None]
+        int_view = {torch.bfloat16: torch.int16, torch.float16: torch.int16,
+                    torch.float32: torch.int32, torch.float64: torch.int64}
+        digest = sum(
+            p.detach().reshape(-1).view(int_view[p.dtype]).long().sum()
+            for p in self.model.parameters()
+        )[None]
         gathered = [torch.zeros_like(digest) for _ in range(self.ctx.dp)]
         dist.all_gather(gathered, digest, group=self.ctx.dp_group)
         if any(int(g[0]) != int(gathered[0][0]) for g in gathered):
             raise RuntimeError(
-                "DP replicas diverged after optimizer step: master "
+                "DP replicas diverged after optimizer step: parameter "
                 f"checksums {[int(g[0]) for g in gathered]} "
                 f"(dp_rank {self.ctx.dp_rank})"
             )

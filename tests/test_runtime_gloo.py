@@ -225,7 +225,7 @@ def _dp_sync_check_worker(rank, world, port, out):
     # inject a divergence on rank 1: the checker must raise on ALL ranks
     if rank == 1:
         with torch.no_grad():
-            runner.optimizer.master[0] += 1.0
+            next(runner.model.parameters()).view(-1)[0] += 1.0
     try:
         runner.verify_replicas_synced()
         out.put(("no-raise", rank))
@@ -245,3 +245,37 @@ def test_dp_sync_checker_detects_divergence():
         p.join(timeout=180)
     results = [out.get() for _ in range(2)]
     assert all(r[0] == "raised" for r in results), results
+
+
+def _zero1_worker(rank, world, port, out):
+    _env(rank, world, port)
+    os.environ["METIS_CHECK_SYNC"] = "1"
+    from metis_amd.runtime.comm import init_parallel
+    from metis_amd.runtime.runner import PlanRunner
+
+    ctx = init_parallel(dp=2, tp=1, pp=1)
+    torch.manual_seed(3)
+    plain = PlanRunner(SPEC, ctx, mbs=2, gbs=8, dtype=torch.float32)
+    torch.manual_seed(3)
+    zero = PlanRunner(SPEC, ctx, mbs=2, gbs=8, dtype=torch.float32, zero1=True)
+
+    # optimizer state really is sharded
+    assert zero.optimizer.master.numel() * 2 == plain.optimizer.master.numel()
+
+    for runner in (plain, zero):
+        torch.manual_seed(100 + rank)
+        runner._data_gen = None
+    l1 = plain.train_step()
+    l2 = zero.train_step()
+    assert abs(l1 - l2) < 1e-6, (l1, l2)
+
+    # identical element-wise update math -> bitwise-equal weights
+    for p1, p2 in zip(plain.model.parameters(), zero.model.parameters()):
+        assert torch.equal(p1, p2)
+    zero.verify_replicas_synced()
+    out.put(("ok", rank))
+    dist.destroy_process_group()
+
+
+def test_zero1_matches_plain_dp():
+    _run_workers(_zero1_worker, port=29617)
