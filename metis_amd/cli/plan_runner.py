@@ -55,7 +55,13 @@ def main() -> None:
                    help="per-block activation recomputation")
     p.add_argument("--zero1", action="store_true",
                    help="shard optimizer state over the DP group")
+    p.add_argument("--layer-partition", default=None,
+                   help='non-uniform stage boundaries from the hetero '
+                        'planner, e.g. "0,13,34" (cumulative, profile '
+                        'layer convention); default: uniform split')
     args = p.parse_args()
+    layer_partition = ([int(x) for x in args.layer_partition.split(",")]
+                       if args.layer_partition else None)
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     spec = MODEL_SPECS[args.model]
@@ -65,7 +71,11 @@ def main() -> None:
             print(f"skip dp{dp}_tp{tp}_pp{pp}: needs world {dp * tp * pp}, have {world}")
             continue
         ctx = init_parallel(dp=dp, tp=tp, pp=pp)
+        if layer_partition is not None:
+            assert len(layer_partition) == pp + 1, (
+                "--layer-partition needs pp+1 boundaries")
         runner = PlanRunner(spec, ctx, mbs=mbs, gbs=gbs,
+                            layer_partition=layer_partition,
                             schedule=args.schedule,
                             recompute=args.recompute,
                             zero1=args.zero1)

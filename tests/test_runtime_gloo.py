@@ -279,3 +279,30 @@ def _zero1_worker(rank, world, port, out):
 
 def test_zero1_matches_plain_dp():
     _run_workers(_zero1_worker, port=29617)
+
+
+def _hetero_partition_worker(rank, world, port, out):
+    _env(rank, world, port)
+    import sys
+
+    from metis_amd.cli import plan_runner
+
+    sys.argv = ["plan_runner", "--model", "gpt2-small", "--plans",
+                "1,1,2,1,2", "--steps", "1", "--warmup", "0",
+                "--layer-partition", "0,1,4",
+                "--out", os.environ["PLAN_OUT"]]
+    plan_runner.MODEL_SPECS = dict(plan_runner.MODEL_SPECS)
+    plan_runner.MODEL_SPECS["gpt2-small"] = SPEC  # 4 profile layers
+    plan_runner.main()
+    out.put(("ok", rank))
+
+
+def test_plan_runner_nonuniform_layer_partition(tmp_path):
+    """The hetero planner's non-uniform stage boundaries execute: stage 0
+    gets 1 profile layer (embedding), stage 1 gets 3 (blocks + head)."""
+    os.environ["PLAN_OUT"] = str(tmp_path / "het.json")
+    _run_workers(_hetero_partition_worker, port=29618)
+    import json as _json
+
+    doc = _json.loads((tmp_path / "het.json").read_text())
+    assert doc["runs"][0]["measured_ms"] > 0
