@@ -321,6 +321,10 @@ class PlanRunner:
                      "mbs": self.mbs, "gbs": self.gbs,
                      "layer_partition": self.layer_partition},
             "rank": self.ctx.rank,
+            # synthetic-data generator state: resume replays the exact
+            # token stream a continuous run would have drawn
+            "data_gen": (self._data_gen.get_state()
+                         if self._data_gen is not None else None),
         }, path)
 
     def load_checkpoint(self, path: str) -> None:
@@ -330,6 +334,10 @@ class PlanRunner:
             self.ctx.dp, self.ctx.tp, self.ctx.pp
         ), "checkpoint plan does not match the running plan"
         self.model.load_state_dict(state["model"])
+        if state.get("data_gen") is not None:
+            dev = self.ctx.device or torch.device("cpu")
+            self._data_gen = torch.Generator(device=dev)
+            self._data_gen.set_state(state["data_gen"])
         opt_state = state["optimizer"]
         self.optimizer.load_state_dict({
             "step": opt_state["step"],
