@@ -36,6 +36,9 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--comm_model", choices=["parity", "alpha_beta"], default="parity")
     p.add_argument("--alpha_us", type=float, default=20.0)
     p.add_argument("--activation_dtype_bytes", type=int, default=1)
+    p.add_argument("--drop_incomplete_partitions", action="store_true",
+                   help="drop plans whose layer partition misses layers "
+                        "(reference balancer quirk; under-costed, unrunnable)")
     p.add_argument("--evaluation_data_path", default=None,
                    help="measured-runtime JSON for cost-model validation")
     p.add_argument("--top_k", type=int, default=0, help="print only the top K plans (0 = all)")
@@ -61,5 +64,6 @@ def parse(argv: Optional[List[str]] = None):
         comm_model=args.comm_model,
         alpha_us=args.alpha_us,
         activation_dtype_bytes=args.activation_dtype_bytes,
+        drop_incomplete_partitions=args.drop_incomplete_partitions,
     )
     return args, model_config, planner_args

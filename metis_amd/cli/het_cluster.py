@@ -71,6 +71,13 @@ def search_het_cluster(
                 except KeyError as e:
                     log.debug("skipping unprofiled plan: %s", e)
                     continue
+                if (planner_args.drop_incomplete_partitions
+                        and intra_plan.layer_partition[-1] != model_config.num_layers):
+                    # reference balancer quirk: a layer fell out of the
+                    # partition; under-costed and unrunnable (config.py)
+                    log.debug("dropping incomplete partition %s",
+                              intra_plan.layer_partition)
+                    continue
                 results.append(
                     (
                         tuple(inter_plan.node_sequence),

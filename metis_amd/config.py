@@ -55,3 +55,9 @@ class PlannerArgs:
     comm_model: str = "parity"
     alpha_us: float = 20.0          # per-collective latency when comm_model="alpha_beta"
     activation_dtype_bytes: int = 1  # 1 => element-count parity (quirk Q8); 2 for bf16 bytes
+    # The reference's LayerComputeBalancer can emit partitions that do NOT
+    # cover every layer on skewed many-stage inputs (slice rounding drops a
+    # layer; reproduced live, e.g. 9 stages x 10 layers -> [..., 8, 9]).
+    # Such plans are under-costed (the missing layer costs nothing) and
+    # unrunnable. Default keeps them for plan-table parity; True drops them.
+    drop_incomplete_partitions: bool = False
