@@ -26,7 +26,10 @@ import torch.nn.functional as F
 from metis_amd.ops import LayerNorm
 from metis_amd.ops.attention import flash_attention
 from metis_amd.ops.cross_entropy import cross_entropy
+from metis_amd.ops.mlp import fused_mlp
 from metis_amd.ops.relayout import heads_merge, qkv_split_transpose
+
+import os as _os
 
 
 @dataclass(frozen=True)
@@ -194,6 +197,10 @@ class GPTBlock(nn.Module):
         self.ln_mlp = LayerNorm(h)
         self.fc1 = ColumnParallelLinear(h, spec.ffn, tp, dtype)
         self.fc2 = RowParallelLinear(spec.ffn, h, tp, dtype)
+        # hipBLASLt epilogue-fused MLP (ops/mlp.py); opt-in until the
+        # GPU numerics of the GELU_AUX_BIAS/DGELU_BGRAD path are validated
+        self._lt_mlp = _os.environ.get("METIS_FC1_EPILOGUE") == "1"
+
 
     def forward(self, x: torch.Tensor, tp_group) -> torch.Tensor:
         residual = x
@@ -208,9 +215,17 @@ class GPTBlock(nn.Module):
 
         residual = x
         y = self.ln_mlp(x)
-        y = self.fc1(y, tp_group)
-        y = F.gelu(y, approximate="tanh")
-        x = residual + self.fc2(y, tp_group)
+        if self._lt_mlp:
+            b, s, hh = y.shape
+            flat = _CopyToTP.apply(y, tp_group).reshape(-1, hh)
+            part = fused_mlp(flat, self.fc1.weight, self.fc1.bias,
+                             self.fc2.weight)
+            part = _ReduceFromTP.apply(part.reshape(b, s, hh), tp_group)
+            x = residual + part + self.fc2.bias
+        else:
+            y = self.fc1(y, tp_group)
+            y = F.gelu(y, approximate="tanh")
+            x = residual + self.fc2(y, tp_group)
         return x
 
 

@@ -13,6 +13,10 @@ void adamw_step(
     torch::Tensor m, torch::Tensor v,
     double lr, double beta1, double beta2, double eps,
     double weight_decay, long step, double grad_scale);
+std::vector<torch::Tensor> lt_fc1_forward(
+    torch::Tensor x, torch::Tensor w, torch::Tensor bias);
+std::vector<torch::Tensor> lt_matmul_dgelu_bgrad(
+    torch::Tensor dy, torch::Tensor w2, torch::Tensor pre_gelu);
 std::vector<torch::Tensor> attn_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v, double scale);
 std::vector<torch::Tensor> attn_bwd(
@@ -75,6 +79,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "dq/dk/dv -> fused dqkv layout");
     m.def("heads_merge", &heads_merge, "[B,H,S,D] -> [B,S,H*D]");
     m.def("heads_unmerge", &heads_unmerge, "[B,S,H*D] -> [B,H,S,D]");
+    m.def("lt_fc1_forward", &lt_fc1_forward,
+          "hipblaslt GEMM with GELU_AUX_BIAS epilogue (fc1 fused)");
+    m.def("lt_matmul_dgelu_bgrad", &lt_matmul_dgelu_bgrad,
+          "hipblaslt GEMM with DGELU_BGRAD epilogue (fc2 data grad fused)");
     m.def("tr16_probe", &tr16_probe,
           "ds_read_b64_tr_b16 lane-semantics probe (debug)");
 }

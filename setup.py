@@ -29,6 +29,7 @@ sources = [
     os.path.join(HIP_DIR, "swiglu.hip"),
     os.path.join(HIP_DIR, "cross_entropy.hip"),
     os.path.join(HIP_DIR, "relayout.hip"),
+    os.path.join(HIP_DIR, "lt_gemm.cpp"),
 ]
 sources = [s for s in sources if os.path.exists(s)]
 
@@ -38,6 +39,7 @@ setup(
         CUDAExtension(
             name="metis_amd._hip_ops",
             sources=sources,
+            libraries=["hipblaslt"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17"],
