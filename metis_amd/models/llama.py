@@ -21,7 +21,10 @@ from metis_amd.models.gpt import (
 from metis_amd.ops.attention import flash_attention
 from metis_amd.ops.cross_entropy import cross_entropy
 from metis_amd.ops.norms import RMSNorm, apply_rope, swiglu
-from metis_amd.ops.relayout import heads_merge, qkv_split_transpose
+from metis_amd.ops.relayout import (heads_merge, qkv_rope_split,
+                                    qkv_split_transpose)
+
+import os as _os
 
 
 @dataclass(frozen=True)
@@ -83,6 +86,10 @@ class LlamaBlock(nn.Module):
         self.gate_up = ColumnParallelLinear(h, 2 * spec.ffn_hidden_size, tp, dtype)
         self.down = RowParallelLinear(spec.ffn_hidden_size, h, tp, dtype)
         self.ffn_per_rank = spec.ffn_hidden_size // tp
+        # fused relayout+RoPE kernel (qkv_rope.hip); opt-in until
+        # GPU-validated (CPU path is identical composition either way)
+        self._fused_qkv_rope = _os.environ.get("METIS_QKV_ROPE") == "1"
+
 
     def forward(self, x: torch.Tensor, tp_group) -> torch.Tensor:
         hq, hkv, d = self.heads_per_rank, self.kv_heads_per_rank, self.head_dim
@@ -90,9 +97,12 @@ class LlamaBlock(nn.Module):
         residual = x
         y = self.norm_attn(x)
         qkv = self.qkv(y, tp_group)
-        q, k, v = qkv_split_transpose(qkv, hq, hkv, d)
-        q = apply_rope(q, self.rope_base)
-        k = apply_rope(k, self.rope_base)
+        if self._fused_qkv_rope:
+            q, k, v = qkv_rope_split(qkv, hq, hkv, d, self.rope_base)
+        else:
+            q, k, v = qkv_split_transpose(qkv, hq, hkv, d)
+            q = apply_rope(q, self.rope_base)
+            k = apply_rope(k, self.rope_base)
         attn = flash_attention(q, k, v, causal=True)
         x = residual + self.proj(heads_merge(attn), tp_group)
 

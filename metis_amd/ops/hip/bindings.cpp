@@ -13,6 +13,12 @@ void adamw_step(
     torch::Tensor m, torch::Tensor v,
     double lr, double beta1, double beta2, double eps,
     double weight_decay, long step, double grad_scale);
+std::vector<torch::Tensor> qkv_rope_split(
+    torch::Tensor qkv, long nq, long nkv, long head_dim,
+    torch::Tensor cos_t, torch::Tensor sin_t);
+torch::Tensor qkv_rope_split_bwd(
+    torch::Tensor dq, torch::Tensor dk, torch::Tensor dv, long head_dim,
+    torch::Tensor cos_t, torch::Tensor sin_t);
 std::vector<torch::Tensor> lt_fc1_forward(
     torch::Tensor x, torch::Tensor w, torch::Tensor bias);
 std::vector<torch::Tensor> lt_matmul_dgelu_bgrad(
@@ -79,6 +85,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "dq/dk/dv -> fused dqkv layout");
     m.def("heads_merge", &heads_merge, "[B,H,S,D] -> [B,S,H*D]");
     m.def("heads_unmerge", &heads_unmerge, "[B,S,H*D] -> [B,H,S,D]");
+    m.def("qkv_rope_split", &qkv_rope_split,
+          "fused QKV relayout + RoPE (neox) forward");
+    m.def("qkv_rope_split_bwd", &qkv_rope_split_bwd,
+          "fused QKV relayout + RoPE backward gather");
     m.def("lt_fc1_forward", &lt_fc1_forward,
           "hipblaslt GEMM with GELU_AUX_BIAS epilogue (fc1 fused)");
     m.def("lt_matmul_dgelu_bgrad", &lt_matmul_dgelu_bgrad,

@@ -63,3 +63,37 @@ def heads_merge(x: torch.Tensor) -> torch.Tensor:
         return _HeadsMerge.apply(x)
     b, h, s, d = x.shape
     return x.transpose(1, 2).reshape(b, s, h * d)
+
+
+# --- fused QKV relayout + RoPE (Llama path; METIS_QKV_ROPE=1) -------------
+class _QKVRoPEFn(torch.autograd.Function):
+    """One pass for qkv_split_transpose + rope(q) + rope(k) (qkv_rope.hip);
+    backward gathers dq/dk/dv into dqkv with the inverse rotation."""
+
+    @staticmethod
+    def forward(ctx, qkv, nq, nkv, d, cos_t, sin_t):
+        ext = _ops.require_extension()
+        q, k, v = ext.qkv_rope_split(qkv, nq, nkv, d, cos_t, sin_t)
+        ctx.save_for_backward(cos_t, sin_t)
+        ctx.d = d
+        return q, k, v
+
+    @staticmethod
+    def backward(ctx, dq, dk, dv):
+        ext = _ops.require_extension()
+        cos_t, sin_t = ctx.saved_tensors
+        dqkv = ext.qkv_rope_split_bwd(
+            dq.contiguous(), dk.contiguous(), dv.contiguous(), ctx.d,
+            cos_t, sin_t)
+        return dqkv, None, None, None, None, None
+
+
+def qkv_rope_split(qkv, nq, nkv, d, base):
+    """[B, S, (nq+2nkv)*d] -> rope(q) [B,nq,S,d], rope(k), v [B,nkv,S,d]."""
+    from metis_amd.ops.norms import apply_rope, rope_tables
+
+    if qkv.is_cuda and qkv.dtype == torch.bfloat16 and d % 16 == 0:
+        cos_t, sin_t = rope_tables(qkv.size(1), d, base, qkv.device)
+        return _QKVRoPEFn.apply(qkv.contiguous(), nq, nkv, d, cos_t, sin_t)
+    q, k, v = qkv_split_transpose(qkv, nq, nkv, d)
+    return apply_rope(q, base), apply_rope(k, base), v

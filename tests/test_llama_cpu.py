@@ -61,3 +61,38 @@ def test_swiglu_cpu():
     a = torch.randn(32)
     b = torch.randn(32)
     assert torch.allclose(swiglu(a, b), torch.nn.functional.silu(a) * b)
+
+
+def test_fused_qkv_rope_cpu_composition():
+    """qkv_rope_split's CPU path must equal split + rope composition."""
+    import torch
+
+    from metis_amd.ops.norms import apply_rope
+    from metis_amd.ops.relayout import qkv_rope_split, qkv_split_transpose
+
+    torch.manual_seed(0)
+    B, S, nq, nkv, d = 2, 16, 4, 2, 32
+    qkv = torch.randn(B, S, (nq + 2 * nkv) * d)
+    q1, k1, v1 = qkv_rope_split(qkv, nq, nkv, d, base=10000.0)
+    q0, k0, v0 = qkv_split_transpose(qkv, nq, nkv, d)
+    assert torch.equal(q1, apply_rope(q0, 10000.0))
+    assert torch.equal(k1, apply_rope(k0, 10000.0))
+    assert torch.equal(v1, v0)
+
+
+def test_llama_block_env_paths_match(monkeypatch):
+    import torch
+
+    from metis_amd.models.llama import LlamaModel, LLAMA_SPECS
+
+    spec = LLAMA_SPECS["llama-tiny"]
+    tok = torch.randint(0, spec.vocab_size, (2, 16))
+    monkeypatch.delenv("METIS_QKV_ROPE", raising=False)
+    torch.manual_seed(0)
+    m1 = LlamaModel(spec, dtype=torch.float32)
+    monkeypatch.setenv("METIS_QKV_ROPE", "1")
+    torch.manual_seed(0)
+    m2 = LlamaModel(spec, dtype=torch.float32)
+    y1 = m1(tok)
+    y2 = m2(tok)
+    assert torch.equal(y1, y2)

@@ -1,5 +1,7 @@
 """GPU numerics tests: each HIP kernel vs a plain PyTorch fp32 reference."""
 
+import os
+
 import pytest
 import torch
 
@@ -172,3 +174,32 @@ def test_relayout_kernels_match_eager():
     gy = torch.randn_like(y)
     y.backward(gy)
     assert torch.equal(x.grad, gy.view(b, s, nq, d).transpose(1, 2).contiguous())
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(os.environ.get("METIS_EXPERIMENTAL") != "1",
+                    reason="fused qkv+rope kernel pending GPU validation")
+def test_qkv_rope_split_gpu_matches_composition():
+    import metis_amd._hip_ops as ext
+    from metis_amd.ops.norms import rope_tables
+
+    torch.manual_seed(0)
+    B, S, nq, nkv, d = 2, 128, 8, 2, 64
+    qkv = torch.randn(B, S, (nq + 2 * nkv) * d, device="cuda",
+                      dtype=torch.bfloat16)
+    cos_t, sin_t = rope_tables(S, d, 500000.0, qkv.device)
+    q1, k1, v1 = ext.qkv_rope_split(qkv, nq, nkv, d, cos_t, sin_t)
+    q0, k0, v0 = ext.qkv_split_transpose(qkv, nq, nkv, d)
+    q0 = ext.rope_apply(q0, cos_t, sin_t, False)
+    k0 = ext.rope_apply(k0, cos_t, sin_t, False)
+    assert torch.allclose(q1.float(), q0.float(), atol=2e-2)
+    assert torch.allclose(k1.float(), k0.float(), atol=2e-2)
+    assert torch.equal(v1, v0)
+
+    # backward gather inverts: dqkv of (dq,dk,dv) rotated back
+    dq, dk, dv = torch.randn_like(q1), torch.randn_like(k1), torch.randn_like(v1)
+    dqkv = ext.qkv_rope_split_bwd(dq, dk, dv, d, cos_t, sin_t)
+    dq0 = ext.rope_apply(dq, cos_t, sin_t, True)
+    dk0 = ext.rope_apply(dk, cos_t, sin_t, True)
+    ref = ext.qkv_split_transpose_bwd(dq0, dk0, dv, d)
+    assert torch.allclose(dqkv.float(), ref.float(), atol=2e-2)
