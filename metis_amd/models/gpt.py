@@ -145,6 +145,62 @@ class _VocabParallelCrossEntropy(torch.autograd.Function):
         return softmax * (grad_out / n), None, None
 
 
+class _VocabParallelCrossEntropyBF16(torch.autograd.Function):
+    """tp>1 cross entropy streaming the bf16 logits shard through the
+    ce_row_max / ce_row_sumexp / cross_entropy_bwd kernels: the fp32
+    [T, V/tp] copy is never materialized. Same Megatron flow (three small
+    all-reduces) as _VocabParallelCrossEntropy; opt-in via METIS_VP_CE=1
+    until GPU-validated."""
+
+    @staticmethod
+    def forward(ctx, logits_shard, labels, group):
+        from metis_amd import ops as _mops
+
+        ext = _mops.require_extension()
+        rank = dist.get_rank(group)
+        vp = logits_shard.size(-1)
+        vocab_start = rank * vp
+
+        m = ext.ce_row_max(logits_shard)
+        dist.all_reduce(m, op=dist.ReduceOp.MAX, group=group)
+        sumexp = ext.ce_row_sumexp(logits_shard, m)
+        dist.all_reduce(sumexp, group=group)
+
+        local = (labels >= vocab_start) & (labels < vocab_start + vp)
+        idx = (labels - vocab_start).clamp(0, vp - 1)
+        target = torch.where(
+            local, logits_shard.gather(1, idx[:, None]).squeeze(1).float(),
+            torch.zeros_like(m))
+        dist.all_reduce(target, group=group)
+
+        loss = (torch.log(sumexp) + m - target).mean()
+        shifted = torch.where(local, idx, torch.full_like(idx, -1))
+        ctx.save_for_backward(logits_shard, shifted, m, sumexp)
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        from metis_amd import ops as _mops
+
+        ext = _mops.require_extension()
+        logits, shifted, m, sumexp = ctx.saved_tensors
+        lse = m + torch.log(sumexp)
+        scale = (grad_out.float() / logits.size(0)).reshape(1)
+        dlogits = ext.cross_entropy_bwd(logits, shifted, lse, scale)
+        return dlogits, None, None
+
+
+def vocab_parallel_ce(logits2d: torch.Tensor, labels: torch.Tensor,
+                      group) -> torch.Tensor:
+    """Vocab-sharded CE; bf16-streaming kernel path under METIS_VP_CE=1."""
+    if (logits2d.is_cuda and logits2d.dtype == torch.bfloat16
+            and _os.environ.get("METIS_VP_CE") == "1"):
+        return _VocabParallelCrossEntropyBF16.apply(
+            logits2d.contiguous(), labels, group)
+    return _VocabParallelCrossEntropy.apply(
+        logits2d.float().contiguous(), labels, group)
+
+
 def _init_linear(weight: torch.Tensor, fan_in: int) -> None:
     std = 1.0 / math.sqrt(fan_in)
     nn.init.normal_(weight, mean=0.0, std=std)
@@ -307,8 +363,8 @@ class GPTModel(nn.Module):
     def _loss(self, logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
         labels = labels.reshape(-1)
         if self.tp_group is not None and dist.get_world_size(self.tp_group) > 1:
-            flat = logits.float().view(-1, logits.size(-1))
-            return _VocabParallelCrossEntropy.apply(flat, labels, self.tp_group)
+            return vocab_parallel_ce(logits.view(-1, logits.size(-1)),
+                                     labels, self.tp_group)
         return cross_entropy(logits.view(-1, logits.size(-1)), labels)
 
     def layer_parameter_bytes(self) -> List[float]:

@@ -16,7 +16,7 @@ import torch.nn as nn
 from metis_amd.models.gpt import (
     ColumnParallelLinear,
     RowParallelLinear,
-    _VocabParallelCrossEntropy,
+    vocab_parallel_ce,
 )
 from metis_amd.ops.attention import flash_attention
 from metis_amd.ops.cross_entropy import cross_entropy
@@ -168,8 +168,8 @@ class LlamaModel(nn.Module):
             if labels is not None:
                 labels = labels.reshape(-1)
                 if self.tp_group is not None and dist.get_world_size(self.tp_group) > 1:
-                    flat = logits.float().view(-1, logits.size(-1))
-                    return _VocabParallelCrossEntropy.apply(flat, labels, self.tp_group)
+                    return vocab_parallel_ce(logits.view(-1, logits.size(-1)),
+                                             labels, self.tp_group)
                 return cross_entropy(logits.view(-1, logits.size(-1)), labels)
             return logits
         return x

@@ -39,7 +39,7 @@ from metis_amd.models.gpt import (
     RowParallelLinear,
     _CopyToTP,
     _ReduceFromTP,
-    _VocabParallelCrossEntropy,
+    vocab_parallel_ce,
     LayerNorm,
     _init_linear,
 )
@@ -273,8 +273,8 @@ class MoEModel(nn.Module):
             if labels is not None:
                 labels = labels.reshape(-1)
                 if self.tp_group is not None and dist.get_world_size(self.tp_group) > 1:
-                    flat = logits.float().view(-1, logits.size(-1))
-                    ce = _VocabParallelCrossEntropy.apply(flat, labels, self.tp_group)
+                    ce = vocab_parallel_ce(logits.view(-1, logits.size(-1)),
+                                           labels, self.tp_group)
                 else:
                     ce = cross_entropy(logits.view(-1, logits.size(-1)), labels)
                 aux = self.consume_aux_loss()

@@ -94,7 +94,90 @@ __global__ void ce_bwd_kernel(
     }
 }
 
+// --- vocab-parallel building blocks --------------------------------------
+// The tp>1 cross entropy keeps its three small all-reduces on the host
+// (Megatron flow) but streams the bf16 shard with these row kernels
+// instead of materializing a fp32 [T, V/tp] copy; its backward reuses
+// ce_bwd_kernel with lse = m_global + log(sumexp_global) and labels
+// shifted to the shard (or -1 when another rank owns the target).
+
+__global__ void row_max_kernel(
+    const bf16x8* __restrict__ logits, float* __restrict__ out,
+    int rows, int vv) {
+    __shared__ float scratch[BLOCK / WAVE_SIZE];
+    for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+        const bf16x8* lrow = logits + (long)row * vv;
+        float m = -1e30f;
+        for (int i = threadIdx.x; i < vv; i += BLOCK) {
+            bf16x8 v = lrow[i];
+            #pragma unroll
+            for (int k = 0; k < VEC; ++k)
+                m = fmaxf(m, bf16_bits_to_float(v[k]));
+        }
+        m = wave_reduce_max(m);
+        const int wv = threadIdx.x / WAVE_SIZE;
+        if ((threadIdx.x & 63) == 0) scratch[wv] = m;
+        __syncthreads();
+        float bm = (threadIdx.x < BLOCK / WAVE_SIZE) ? scratch[threadIdx.x]
+                                                     : -1e30f;
+        bm = wave_reduce_max(bm);
+        if (threadIdx.x == 0) out[row] = bm;
+        __syncthreads();
+    }
+}
+
+__global__ void row_sumexp_kernel(
+    const bf16x8* __restrict__ logits, const float* __restrict__ m,
+    float* __restrict__ out, int rows, int vv) {
+    __shared__ float scratch[BLOCK / WAVE_SIZE];
+    for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+        const bf16x8* lrow = logits + (long)row * vv;
+        const float mr = m[row];
+        float sum = 0.f;
+        for (int i = threadIdx.x; i < vv; i += BLOCK) {
+            bf16x8 v = lrow[i];
+            #pragma unroll
+            for (int k = 0; k < VEC; ++k)
+                sum += __expf(bf16_bits_to_float(v[k]) - mr);
+        }
+        sum = block_reduce_sum(sum, scratch);
+        if (threadIdx.x == 0) out[row] = sum;
+        __syncthreads();
+    }
+}
+
 }  // namespace
+
+torch::Tensor ce_row_max(torch::Tensor logits) {
+    TORCH_CHECK(logits.is_cuda() && logits.dtype() == torch::kBFloat16);
+    TORCH_CHECK(logits.dim() == 2 && logits.size(1) % 8 == 0);
+    auto lc = logits.contiguous();
+    const long N = logits.size(0), V = logits.size(1);
+    auto out = torch::empty({N}, logits.options().dtype(torch::kFloat32));
+    const int grid = (int)std::min<long>(N, 2048);
+    hipLaunchKernelGGL(row_max_kernel, dim3(grid), dim3(BLOCK), 0,
+        c10::hip::getCurrentHIPStream().stream(),
+        reinterpret_cast<const bf16x8*>(lc.data_ptr()),
+        out.data_ptr<float>(), (int)N, (int)(V / 8));
+    HIP_CHECK_LAST();
+    return out;
+}
+
+torch::Tensor ce_row_sumexp(torch::Tensor logits, torch::Tensor m) {
+    TORCH_CHECK(logits.is_cuda() && logits.dtype() == torch::kBFloat16);
+    auto lc = logits.contiguous();
+    const long N = logits.size(0), V = logits.size(1);
+    TORCH_CHECK(m.size(0) == N && m.dtype() == torch::kFloat32);
+    auto out = torch::empty({N}, logits.options().dtype(torch::kFloat32));
+    const int grid = (int)std::min<long>(N, 2048);
+    hipLaunchKernelGGL(row_sumexp_kernel, dim3(grid), dim3(BLOCK), 0,
+        c10::hip::getCurrentHIPStream().stream(),
+        reinterpret_cast<const bf16x8*>(lc.data_ptr()),
+        m.contiguous().data_ptr<float>(),
+        out.data_ptr<float>(), (int)N, (int)(V / 8));
+    HIP_CHECK_LAST();
+    return out;
+}
 
 std::vector<torch::Tensor> cross_entropy_fwd(
     torch::Tensor logits, torch::Tensor labels) {

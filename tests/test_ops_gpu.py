@@ -203,3 +203,31 @@ def test_qkv_rope_split_gpu_matches_composition():
     dk0 = ext.rope_apply(dk, cos_t, sin_t, True)
     ref = ext.qkv_split_transpose_bwd(dq0, dk0, dv, d)
     assert torch.allclose(dqkv.float(), ref.float(), atol=2e-2)
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(os.environ.get("METIS_EXPERIMENTAL") != "1",
+                    reason="vocab-parallel CE kernels pending GPU validation")
+def test_ce_row_kernels_match_torch():
+    import metis_amd._hip_ops as ext
+
+    torch.manual_seed(0)
+    x = torch.randn(512, 6400, device="cuda", dtype=torch.bfloat16) * 4
+    m = ext.ce_row_max(x)
+    assert torch.allclose(m, x.float().max(dim=1).values, atol=1e-6)
+    se = ext.ce_row_sumexp(x, m)
+    ref = torch.exp(x.float() - m[:, None]).sum(dim=1)
+    assert torch.allclose(se, ref, rtol=1e-3)
+
+    # bwd reuse with shifted labels: -1 (unowned) must add no onehot
+    labels = torch.randint(0, 6400, (512,), device="cuda")
+    labels[::2] = -1
+    lse = m + torch.log(se)
+    scale = torch.tensor([0.5 / 512], device="cuda")
+    d = ext.cross_entropy_bwd(x, labels, lse, scale)
+    soft = torch.softmax(x.float(), dim=1)
+    oh = torch.zeros_like(soft)
+    owned = labels >= 0
+    oh[owned] = torch.nn.functional.one_hot(labels[owned], 6400).float()
+    ref_d = (soft - oh) * float(scale)
+    assert torch.allclose(d.float(), ref_d, atol=5e-3)
