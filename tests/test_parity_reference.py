@@ -309,3 +309,78 @@ def test_het_search_parity_two_device_types(tmp_path):
     theirs = ref_rows
     assert len(mine) == len(theirs), (len(mine), len(theirs))
     assert mine == theirs
+
+
+def test_het_search_parity_randomized_configs(tmp_path):
+    """Randomized-config live parity: single-type clusters with varied
+    gbs / node count / slots / bandwidth / synthetic profile speeds, full
+    search compared cost-and-plan-exact against the reference CLI."""
+    import importlib.util
+
+    spec = importlib.util.spec_from_file_location(
+        "gen_synth2", os.path.join(REPO, "scripts", "gen_synth_profiles.py"))
+    gen = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(gen)
+
+    from metis_amd.cluster import ClusterSpec
+    from metis_amd.config import ModelConfig, PlannerArgs
+    from metis_amd.cli.het_cluster import search_het_cluster
+    from metis_amd.profiles import ProfileStore
+
+    rng = random.Random(20260913)
+    for case in range(3):
+        prof = tmp_path / f"prof{case}"
+        speed = rng.choice([0.6, 1.0, 1.4])
+        gen.main(str(prof), device_speeds=(("A100", speed),))
+        nodes = rng.choice([2, 4])
+        slots = rng.choice([2, 4])
+        bw = rng.choice([25, 50, 100])
+        gbs = rng.choice([8, 16, 32])
+        hf = tmp_path / f"hostfile{case}"
+        hf.write_text("".join(f"n{i} slots={slots}\n" for i in range(nodes)))
+        cf = tmp_path / f"clusterfile{case}.json"
+        cf.write_text(json.dumps({
+            f"n{i}": {"instance_type": "A100", "inter_bandwidth": bw,
+                      "intra_bandwidth": bw, "memory": 80}
+            for i in range(nodes)
+        }))
+        common = [
+            "--model_name", "GPT", "--num_layers", "10", "--gbs", str(gbs),
+            "--hidden_size", "4096", "--sequence_length", "1024",
+            "--vocab_size", "51200", "--attention_head_size", "128",
+            "--hostfile_path", str(hf), "--clusterfile_path", str(cf),
+            "--profile_data_path", str(prof),
+            "--max_profiled_tp_degree", "4", "--max_profiled_batch_size", "4",
+            "--min_group_scale_variance", "1", "--max_permute_len", "4",
+        ]
+        ref_out = subprocess.run(
+            [sys.executable, "cost_het_cluster.py"] + common,
+            cwd="/root/reference", capture_output=True, text=True, check=True,
+        ).stdout
+        ref_rows = []
+        for line in ref_out.splitlines():
+            parts = line.split(", ")
+            if parts and parts[0].isdigit():
+                rest = line[line.index(", [") + 2:]
+                ref_rows.append((round(float(parts[1]), 8), rest))
+
+        model_from = probe("homo_costs", {
+            "hostfile": str(hf), "clusterfile": str(cf),
+            "profile_dir": str(prof), "gbs": gbs, "max_tp": 4,
+            "model": dict(model_name="GPT", num_layers=10, hidden_size=4096,
+                          sequence_length=1024, vocab_size=51200,
+                          attention_head_size=128),
+        })["model_file_order"][0]
+        cluster = ClusterSpec(str(hf), str(cf))
+        store = ProfileStore.load_dir(str(prof), model_from=model_from)
+        cfg = ModelConfig("GPT", 10, 4096, 1024, 51200, 128)
+        results = search_het_cluster(
+            cluster, store, cfg,
+            PlannerArgs(gbs=gbs, max_profiled_tp_degree=4,
+                        max_profiled_batch_size=4,
+                        min_group_scale_variance=1, max_permute_len=4))
+        mine = sorted(
+            (round(r[6], 8),
+             f"{r[1]}, {r[2]}, {r[3]}, {r[4]}") for r in results)
+        assert mine == sorted(ref_rows), (
+            case, speed, nodes, slots, bw, gbs, len(mine), len(ref_rows))
