@@ -42,6 +42,8 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--evaluation_data_path", default=None,
                    help="measured-runtime JSON for cost-model validation")
     p.add_argument("--top_k", type=int, default=0, help="print only the top K plans (0 = all)")
+    p.add_argument("--json_out", default=None,
+                   help="also write the ranked plans as structured JSON")
     return p
 
 

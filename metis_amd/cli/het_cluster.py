@@ -114,6 +114,20 @@ def main(argv: Optional[List[str]] = None) -> List[PlanResult]:
     for idx, r in enumerate(ranked):
         node_seq = [str(s) for s in r[0]]
         print(f"{idx + 1}, {r[6]}, {node_seq}, {r[1]}, {r[2]}, {r[3]}, {r[4]}")
+    if args.json_out:
+        import json as _json
+
+        with open(args.json_out, "w") as fh:
+            _json.dump({"num_plans": len(results), "plans": [
+                {"rank": i + 1, "cost_ms": r[6],
+                 "node_sequence": [str(s) for s in r[0]],
+                 "device_groups": r[1],
+                 "strategies": [list(st) for st in r[2]],
+                 "batches": r[3], "layer_partition": r[4],
+                 "num_repartition": r[5]}
+                for i, r in enumerate(ranked)
+            ]}, fh, indent=2)
+        print(f"wrote {args.json_out}")
     return ranked
 
 

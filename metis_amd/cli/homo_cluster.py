@@ -71,6 +71,18 @@ def main(argv: Optional[List[str]] = None) -> List[Tuple[UniformPlan, float, boo
         suffix = "  [OOM]" if oom else ""
         print(f"{idx + 1}, {cost}, {plan}{suffix}")
 
+    if args.json_out:
+        import json as _json
+
+        with open(args.json_out, "w") as fh:
+            _json.dump({"num_plans": len(results), "plans": [
+                {"rank": i + 1, "cost_ms": cost, "oom": oom,
+                 "plan": {"dp": p.dp, "tp": p.tp, "pp": p.pp,
+                          "mbs": p.mbs, "gbs": p.gbs}}
+                for i, (p, cost, oom) in enumerate(ranked)
+            ]}, fh, indent=2)
+        print(f"wrote {args.json_out}")
+
     if args.evaluation_data_path:
         validator = CostValidator(args.evaluation_data_path)
         estimates = {
