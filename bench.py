@@ -52,6 +52,8 @@ def parse_args():
     p.add_argument("--recompute", action="store_true")
     p.add_argument("--zero1", action="store_true",
                    help="shard optimizer state over the DP group")
+    p.add_argument("--sp", action="store_true",
+                   help="sequence parallelism in the TP norm regions (GPT)")
     return p.parse_args()
 
 
@@ -94,7 +96,7 @@ def main() -> None:
     ctx = init_parallel(dp=dp, tp=tp, pp=pp)
     runner = PlanRunner(spec, ctx, mbs=mbs, gbs=gbs,
                         schedule=args.schedule, recompute=args.recompute,
-                        zero1=args.zero1)
+                        zero1=args.zero1, sp=args.sp)
 
     ms = runner.timed_steps(args.steps, args.warmup)
 

@@ -55,6 +55,8 @@ def main() -> None:
                    help="per-block activation recomputation")
     p.add_argument("--zero1", action="store_true",
                    help="shard optimizer state over the DP group")
+    p.add_argument("--sp", action="store_true",
+                   help="sequence parallelism in the TP norm regions (GPT)")
     p.add_argument("--layer-partition", default=None,
                    help='non-uniform stage boundaries from the hetero '
                         'planner, e.g. "0,13,34" (cumulative, profile '
@@ -78,7 +80,7 @@ def main() -> None:
                             layer_partition=layer_partition,
                             schedule=args.schedule,
                             recompute=args.recompute,
-                            zero1=args.zero1)
+                            zero1=args.zero1, sp=args.sp)
         ms = runner.timed_steps(args.steps, args.warmup)
         if dist.is_initialized():
             t = torch.tensor([ms], dtype=torch.float64,

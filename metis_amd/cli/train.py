@@ -72,6 +72,7 @@ def main() -> None:
     p.add_argument("--schedule", default="gpipe", choices=("gpipe", "1f1b"))
     p.add_argument("--recompute", action="store_true")
     p.add_argument("--zero1", action="store_true")
+    p.add_argument("--sp", action="store_true")
     p.add_argument("--layer-partition", default=None)
     p.add_argument("--log-every", type=int, default=10)
     args = p.parse_args()
@@ -83,7 +84,7 @@ def main() -> None:
     runner = PlanRunner(MODEL_SPECS[args.model], ctx, mbs=args.mbs,
                         gbs=args.gbs, layer_partition=lp,
                         schedule=args.schedule, recompute=args.recompute,
-                        zero1=args.zero1)
+                        zero1=args.zero1, sp=args.sp)
 
     start = 0
     if args.resume and args.checkpoint_dir:

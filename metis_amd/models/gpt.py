@@ -104,6 +104,79 @@ class _ReduceFromTP(torch.autograd.Function):
         return grad, None
 
 
+# --- sequence-parallel collectives (Megatron-SP; opt-in via sp=True) ------
+# With SP the norm/residual regions hold [b, s/tp, h] instead of the full
+# sequence: entering a column-parallel layer the shard is all-gathered
+# along seq (backward: reduce-scatter), and the row-parallel output is
+# reduce-scattered (backward: all-gather). Same wire bytes as the TP
+# ring all-reduce (ar = rs + ag) but the replicated-region activations
+# shrink by 1/tp. gloo has no reduce-scatter, so CPU tests emulate it
+# with all-reduce + slice (numerically identical).
+
+def _rs_seq(x: torch.Tensor, group) -> torch.Tensor:
+    """reduce-scatter along dim 1: [b, s, h] (partial sums) -> [b, s/ws, h]."""
+    ws = dist.get_world_size(group)
+    r = dist.get_rank(group)
+    sl = x.size(1) // ws
+    if dist.get_backend(group) == "gloo":
+        x = x.contiguous()
+        dist.all_reduce(x, group=group)
+        return x[:, r * sl:(r + 1) * sl].contiguous()
+    # [b, s, h] -> [ws, b, sl, h] so reduce_scatter_tensor splits on dim 0
+    parts = x.reshape(x.size(0), ws, sl, x.size(2)).permute(1, 0, 2, 3).contiguous()
+    out = torch.empty_like(parts[0])
+    dist.reduce_scatter_tensor(out, parts, group=group)
+    return out
+
+
+def _ag_seq(x: torch.Tensor, group) -> torch.Tensor:
+    """all-gather along dim 1: [b, s/ws, h] -> [b, s, h]."""
+    ws = dist.get_world_size(group)
+    x = x.contiguous()
+    if dist.get_backend(group) == "gloo":
+        parts = [torch.empty_like(x) for _ in range(ws)]
+        dist.all_gather(parts, x, group=group)
+        return torch.cat(parts, dim=1)
+    out = torch.empty(ws, x.size(0), x.size(1), x.size(2),
+                      dtype=x.dtype, device=x.device)
+    dist.all_gather_into_tensor(out, x, group=group)
+    return out.permute(1, 0, 2, 3).reshape(x.size(0), ws * x.size(1), x.size(2))
+
+
+class _GatherSeq(torch.autograd.Function):
+    """fwd: seq all-gather; bwd: seq reduce-scatter (SP's f operator)."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        if group is None or dist.get_world_size(group) <= 1:
+            return x
+        return _ag_seq(x, group)
+
+    @staticmethod
+    def backward(ctx, grad):
+        if ctx.group is None or dist.get_world_size(ctx.group) <= 1:
+            return grad, None
+        return _rs_seq(grad, ctx.group), None
+
+
+class _ReduceScatterSeq(torch.autograd.Function):
+    """fwd: seq reduce-scatter; bwd: seq all-gather (SP's g operator)."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        if group is None or dist.get_world_size(group) <= 1:
+            return x
+        return _rs_seq(x, group)
+
+    @staticmethod
+    def backward(ctx, grad):
+        if ctx.group is None or dist.get_world_size(ctx.group) <= 1:
+            return grad, None
+        return _ag_seq(grad.contiguous(), ctx.group), None
+
+
 class _VocabParallelCrossEntropy(torch.autograd.Function):
     """Cross entropy over vocab-sharded logits without gathering them
     (Megatron-style): three small all-reduces (max, sum-exp, target logit)
@@ -258,25 +331,48 @@ class GPTBlock(nn.Module):
         self._lt_mlp = _os.environ.get("METIS_FC1_EPILOGUE") == "1"
 
 
-    def forward(self, x: torch.Tensor, tp_group) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, tp_group, sp: bool = False) -> torch.Tensor:
+        """With ``sp`` the block holds [b, s/tp, h] in the norm/residual
+        regions: the f/g operators become seq all-gather / reduce-scatter
+        (see the SP collectives above); attention always sees the full
+        sequence."""
         residual = x
         y = self.ln_attn(x)
         # per-rank qkv layout: [q heads | k heads | v heads] blocks
-        qkv = self.qkv(y, tp_group)
+        if sp:
+            y_full = _GatherSeq.apply(y, tp_group)
+            qkv = F.linear(y_full, self.qkv.weight, self.qkv.bias)
+        else:
+            qkv = self.qkv(y, tp_group)
         q, k, v = qkv_split_transpose(
             qkv, self.heads_per_rank, self.heads_per_rank, self.head_dim
         )
         attn = flash_attention(q, k, v, causal=True)
-        x = residual + self.proj(heads_merge(attn), tp_group)
+        if sp:
+            part = F.linear(heads_merge(attn), self.proj.weight)
+            part = _ReduceScatterSeq.apply(part, tp_group)
+            x = residual + part + self.proj.bias
+        else:
+            x = residual + self.proj(heads_merge(attn), tp_group)
 
         residual = x
         y = self.ln_mlp(x)
         if self._lt_mlp:
-            b, s, hh = y.shape
-            flat = _CopyToTP.apply(y, tp_group).reshape(-1, hh)
+            inp = (_GatherSeq.apply(y, tp_group) if sp
+                   else _CopyToTP.apply(y, tp_group))
+            b, s, hh = inp.shape
+            flat = inp.reshape(-1, hh)
             part = fused_mlp(flat, self.fc1.weight, self.fc1.bias,
-                             self.fc2.weight)
-            part = _ReduceFromTP.apply(part.reshape(b, s, hh), tp_group)
+                             self.fc2.weight).reshape(b, s, hh)
+            part = (_ReduceScatterSeq.apply(part, tp_group) if sp
+                    else _ReduceFromTP.apply(part, tp_group))
+            x = residual + part + self.fc2.bias
+        elif sp:
+            y_full = _GatherSeq.apply(y, tp_group)
+            t = F.gelu(F.linear(y_full, self.fc1.weight, self.fc1.bias),
+                       approximate="tanh")
+            part = _ReduceScatterSeq.apply(F.linear(t, self.fc2.weight),
+                                           tp_group)
             x = residual + part + self.fc2.bias
         else:
             y = self.fc1(y, tp_group)
@@ -300,11 +396,15 @@ class GPTModel(nn.Module):
         dtype: torch.dtype = torch.bfloat16,
         layer_range: Optional[tuple] = None,
         tp_group=None,
+        sp: bool = False,
     ):
         super().__init__()
         self.spec = spec
         self.tp = tp
         self.tp_group = tp_group
+        self.sp = sp and tp > 1
+        if self.sp:
+            assert spec.seq_length % tp == 0, "sp needs seq % tp == 0"
         total = spec.profile_num_layers
         start, end = layer_range if layer_range is not None else (0, total)
         assert 0 <= start < end <= total
@@ -333,28 +433,69 @@ class GPTModel(nn.Module):
         # compute for O(1) per-block activation memory)
         self.recompute = False
 
+        # SP: params in the seq-sharded regions are replicated but see
+        # only this rank's token slice, so their grads are partial — sum
+        # them across the TP group in backward (same hook pattern as the
+        # MoE router)
+        if self.sp and tp_group is not None:
+            hooked = []
+            if self.has_embedding:
+                hooked += [self.wte.weight, self.wpe.weight]
+            for blk in self.blocks:
+                hooked += [blk.ln_attn.weight, blk.ln_attn.bias,
+                           blk.ln_mlp.weight, blk.ln_mlp.bias,
+                           blk.proj.bias, blk.fc2.bias]
+            if self.has_head:
+                # ln_final runs on the full sequence but its upstream grad
+                # is the pre-reduce-scatter partial (the head bypasses f)
+                hooked += [self.ln_final.weight, self.ln_final.bias]
+
+            def _sync(p):
+                if p.grad is not None:
+                    dist.all_reduce(p.grad, group=tp_group)
+
+            for p in hooked:
+                p.register_post_accumulate_grad_hook(_sync)
+
     def forward(
         self, x: torch.Tensor, labels: Optional[torch.Tensor] = None
     ) -> torch.Tensor:
-        """x: token ids [b, s] on the first stage, hidden states elsewhere.
-        Returns the loss when this stage has the head and labels are given,
-        otherwise the stage's output hidden states."""
+        """x: token ids [b, s] on the first stage, hidden states elsewhere
+        ([b, s/tp, h] between blocks when sp). Returns the loss when this
+        stage has the head and labels are given, otherwise the stage's
+        output hidden states."""
         if self.has_embedding:
             b, s = x.shape
-            pos = torch.arange(s, device=x.device)
+            if self.sp:
+                # embed only this rank's sequence slice
+                r = dist.get_rank(self.tp_group)
+                ss = s // self.tp
+                x = x[:, r * ss:(r + 1) * ss]
+                pos = torch.arange(r * ss, (r + 1) * ss, device=x.device)
+            else:
+                pos = torch.arange(s, device=x.device)
             x = self.wte(x) + self.wpe(pos)[None, :, :]
 
         use_ckpt = self.recompute and torch.is_grad_enabled()
         for block in self.blocks:
             if use_ckpt:
                 x = torch.utils.checkpoint.checkpoint(
-                    block, x, self.tp_group, use_reentrant=False)
+                    block, x, self.tp_group, self.sp, use_reentrant=False)
             else:
-                x = block(x, self.tp_group)
+                x = block(x, self.tp_group, self.sp)
 
         if self.has_head:
-            x = self.ln_final(x)
-            logits = self.head(x, self.tp_group)  # [b, s, vocab/tp]
+            if self.sp:
+                # the head + loss work on the full sequence; the gather
+                # REPLACES the head's f operator (its backward
+                # reduce-scatter does the cross-rank sum — going through
+                # _CopyToTP too would double the gradient)
+                x = _GatherSeq.apply(x, self.tp_group)
+                x = self.ln_final(x)
+                logits = F.linear(x, self.head.weight, self.head.bias)
+            else:
+                x = self.ln_final(x)
+                logits = self.head(x, self.tp_group)  # [b, s, vocab/tp]
             if labels is not None:
                 return self._loss(logits, labels)
             return logits

@@ -43,6 +43,7 @@ class PlanRunner:
         schedule: str = "gpipe",
         recompute: bool = False,
         zero1: bool = False,
+        sp: bool = False,
     ) -> None:
         assert schedule in ("gpipe", "1f1b")
         self.schedule = schedule
@@ -71,9 +72,13 @@ class PlanRunner:
             model_cls = LlamaModel
         else:
             model_cls = GPTModel
+        self.sp = bool(sp) and ctx.tp > 1 and model_cls is GPTModel
+        extra = {"sp": self.sp} if model_cls is GPTModel else {}
+        assert not (sp and not self.sp and ctx.tp > 1), (
+            "sequence parallelism is implemented for the GPT family")
         self.model = model_cls(
             spec, tp=ctx.tp, dtype=dtype, layer_range=(start, end),
-            tp_group=ctx.tp_group,
+            tp_group=ctx.tp_group, **extra,
         )
         self.model.recompute = recompute
         if ctx.device is not None:
@@ -152,7 +157,8 @@ class PlanRunner:
     def _step_pipeline(self) -> float:
         ctx = self.ctx
         h = self.spec.hidden_size
-        act_shape = (self.mbs, self.spec.seq_length, h)
+        seq = self.spec.seq_length // (ctx.tp if self.sp else 1)
+        act_shape = (self.mbs, seq, h)
         prev = ctx.stage_neighbor(-1) if not ctx.is_first_stage else None
         nxt = ctx.stage_neighbor(+1) if not ctx.is_last_stage else None
 
@@ -213,7 +219,8 @@ class PlanRunner:
         deadlock with rendezvous sends); recvs stay blocking."""
         ctx = self.ctx
         nm = self.num_microbatches
-        act_shape = (self.mbs, self.spec.seq_length, self.spec.hidden_size)
+        seq = self.spec.seq_length // (ctx.tp if self.sp else 1)
+        act_shape = (self.mbs, seq, self.spec.hidden_size)
         prev = ctx.stage_neighbor(-1) if not ctx.is_first_stage else None
         nxt = ctx.stage_neighbor(+1) if not ctx.is_last_stage else None
         self.optimizer.zero_grad()
