@@ -161,6 +161,7 @@ class PlanRunner:
         act_shape = (self.mbs, seq, h)
         prev = ctx.stage_neighbor(-1) if not ctx.is_first_stage else None
         nxt = ctx.stage_neighbor(+1) if not ctx.is_last_stage else None
+        dev = self.ctx.device or "cpu"
 
         self.optimizer.zero_grad()
         inputs: List[Optional[torch.Tensor]] = []
@@ -168,14 +169,33 @@ class PlanRunner:
         losses: List[torch.Tensor] = []
         aux_terms: List[Optional[torch.Tensor]] = []
 
+        # Pre-post every fill-phase activation recv (and, in reverse order,
+        # every drain-phase grad recv) so p2p transfers land while this
+        # stage computes. GPipe keeps all nm activations live anyway, so
+        # the buffers cost nothing extra; senders emit in this exact order.
+        nm = self.num_microbatches
+        fwd_bufs, fwd_reqs = [], []
+        if not ctx.is_first_stage:
+            for _ in range(nm):
+                buf = torch.empty(act_shape, dtype=self.dtype, device=dev)
+                fwd_bufs.append(buf)
+                fwd_reqs.append(dist.irecv(buf, src=prev))
+        bwd_bufs, bwd_reqs = {}, {}
+        if not ctx.is_last_stage:
+            for i in reversed(range(nm)):
+                buf = torch.empty(act_shape, dtype=self.dtype, device=dev)
+                bwd_bufs[i] = buf
+                bwd_reqs[i] = dist.irecv(buf, src=nxt)
+
         # forward fill
-        for _ in range(self.num_microbatches):
+        for mb in range(nm):
             if ctx.is_first_stage:
                 tokens, labels = self.synthetic_batch()
                 x = tokens
                 inputs.append(None)
             else:
-                x = self._recv_activation(act_shape, prev).requires_grad_(True)
+                fwd_reqs[mb].wait()
+                x = fwd_bufs[mb].requires_grad_(True)
                 inputs.append(x)
             if ctx.is_last_stage:
                 if ctx.is_first_stage:
@@ -200,8 +220,8 @@ class PlanRunner:
             if ctx.is_last_stage:
                 (outputs[i] / self.num_microbatches).backward()
             else:
-                gout = self._recv_activation(act_shape, nxt)
-                self._backward_stage(outputs[i], gout, aux_terms[i])
+                bwd_reqs[i].wait()
+                self._backward_stage(outputs[i], bwd_bufs[i], aux_terms[i])
             if not ctx.is_first_stage:
                 dist.send(inputs[i].grad.contiguous(), dst=prev)
 
