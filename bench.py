@@ -48,7 +48,9 @@ def parse_args():
                         "--profile-dir instead of the flags")
     p.add_argument("--profile-dir", default=None,
                    help="defaults to profiles/mi355x/<model>")
-    p.add_argument("--schedule", default="gpipe", choices=("gpipe", "1f1b"))
+    p.add_argument("--schedule", default="gpipe", choices=("gpipe", "1f1b", "interleaved"))
+    p.add_argument("--vpp", type=int, default=2,
+                   help="virtual chunks per rank for --schedule interleaved")
     p.add_argument("--recompute", action="store_true")
     p.add_argument("--zero1", action="store_true",
                    help="shard optimizer state over the DP group")
@@ -96,7 +98,7 @@ def main() -> None:
     ctx = init_parallel(dp=dp, tp=tp, pp=pp)
     runner = PlanRunner(spec, ctx, mbs=mbs, gbs=gbs,
                         schedule=args.schedule, recompute=args.recompute,
-                        zero1=args.zero1, sp=args.sp)
+                        zero1=args.zero1, sp=args.sp, vpp=args.vpp)
 
     ms = runner.timed_steps(args.steps, args.warmup)
 

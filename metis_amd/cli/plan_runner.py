@@ -50,7 +50,9 @@ def main() -> None:
     p.add_argument("--steps", type=int, default=5)
     p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--out", default="profiles/measured_runs.json")
-    p.add_argument("--schedule", default="gpipe", choices=("gpipe", "1f1b"))
+    p.add_argument("--schedule", default="gpipe", choices=("gpipe", "1f1b", "interleaved"))
+    p.add_argument("--vpp", type=int, default=2,
+                   help="virtual chunks per rank for --schedule interleaved")
     p.add_argument("--recompute", action="store_true",
                    help="per-block activation recomputation")
     p.add_argument("--zero1", action="store_true",
@@ -80,7 +82,7 @@ def main() -> None:
                             layer_partition=layer_partition,
                             schedule=args.schedule,
                             recompute=args.recompute,
-                            zero1=args.zero1, sp=args.sp)
+                            zero1=args.zero1, sp=args.sp, vpp=args.vpp)
         ms = runner.timed_steps(args.steps, args.warmup)
         if dist.is_initialized():
             t = torch.tensor([ms], dtype=torch.float64,

@@ -69,7 +69,9 @@ def main() -> None:
     p.add_argument("--checkpoint-every", type=int, default=100)
     p.add_argument("--resume", action="store_true",
                    help="continue from the latest COMPLETE checkpoint")
-    p.add_argument("--schedule", default="gpipe", choices=("gpipe", "1f1b"))
+    p.add_argument("--schedule", default="gpipe", choices=("gpipe", "1f1b", "interleaved"))
+    p.add_argument("--vpp", type=int, default=2,
+                   help="virtual chunks per rank for --schedule interleaved")
     p.add_argument("--recompute", action="store_true")
     p.add_argument("--zero1", action="store_true")
     p.add_argument("--sp", action="store_true")
@@ -84,7 +86,7 @@ def main() -> None:
     runner = PlanRunner(MODEL_SPECS[args.model], ctx, mbs=args.mbs,
                         gbs=args.gbs, layer_partition=lp,
                         schedule=args.schedule, recompute=args.recompute,
-                        zero1=args.zero1, sp=args.sp)
+                        zero1=args.zero1, sp=args.sp, vpp=args.vpp)
 
     start = 0
     if args.resume and args.checkpoint_dir:

@@ -44,9 +44,14 @@ class PlanRunner:
         recompute: bool = False,
         zero1: bool = False,
         sp: bool = False,
+        vpp: int = 1,
     ) -> None:
-        assert schedule in ("gpipe", "1f1b")
+        assert schedule in ("gpipe", "1f1b", "interleaved")
         self.schedule = schedule
+        self.vpp = vpp if schedule == "interleaved" else 1
+        if schedule == "interleaved":
+            assert vpp >= 2, "interleaved needs >= 2 virtual chunks (--vpp)"
+            assert ctx.pp > 1, "interleaved needs pp > 1"
         self.spec = spec
         self.ctx = ctx
         self.mbs = mbs
@@ -56,15 +61,19 @@ class PlanRunner:
             "GPipe needs at least pp microbatches"
         )
         assert gbs % (mbs * ctx.dp) == 0, "gbs must divide by mbs*dp"
+        if schedule == "interleaved":
+            assert self.num_microbatches % ctx.pp == 0, (
+                "interleaved schedule needs num_microbatches % pp == 0")
 
         total_layers = spec.profile_num_layers
+        num_stages = ctx.pp * self.vpp
         if layer_partition is None:
-            counts = uniform_layer_split(total_layers, ctx.pp)
+            counts = uniform_layer_split(total_layers, num_stages)
             layer_partition = [0]
             for c in counts:
                 layer_partition.append(layer_partition[-1] + c)
+        assert len(layer_partition) == num_stages + 1
         self.layer_partition = layer_partition
-        start, end = layer_partition[ctx.pp_rank], layer_partition[ctx.pp_rank + 1]
 
         if isinstance(spec, MoEModelSpec):
             model_cls = MoEModel
@@ -76,11 +85,21 @@ class PlanRunner:
         extra = {"sp": self.sp} if model_cls is GPTModel else {}
         assert not (sp and not self.sp and ctx.tp > 1), (
             "sequence parallelism is implemented for the GPT family")
-        self.model = model_cls(
-            spec, tp=ctx.tp, dtype=dtype, layer_range=(start, end),
-            tp_group=ctx.tp_group, **extra,
-        )
-        self.model.recompute = recompute
+
+        # one model slice per virtual chunk: chunk c covers virtual stage
+        # c*pp + pp_rank (vpp == 1 -> the ordinary single slice)
+        chunks = []
+        for c in range(self.vpp):
+            vs = c * ctx.pp + ctx.pp_rank
+            start, end = layer_partition[vs], layer_partition[vs + 1]
+            chunks.append(model_cls(
+                spec, tp=ctx.tp, dtype=dtype, layer_range=(start, end),
+                tp_group=ctx.tp_group, **extra,
+            ))
+        self.model_chunks = chunks
+        self.model = chunks[0] if self.vpp == 1 else torch.nn.ModuleList(chunks)
+        for chunk in chunks:
+            chunk.recompute = recompute
         if ctx.device is not None:
             self.model.to(ctx.device)
         shard_group = ctx.dp_group if (zero1 and ctx.dp > 1) else None
@@ -295,6 +314,105 @@ class PlanRunner:
             return float(torch.stack(losses).mean())
         return 0.0
 
+    def _rank_of_stage(self, p: int) -> int:
+        ctx = self.ctx
+        return (p * ctx.dp + ctx.dp_rank) * ctx.tp + ctx.tp_rank
+
+    def _step_interleaved(self) -> float:
+        """Interleaved 1F1B (virtual pipeline stages): rank r hosts vpp
+        chunks, chunk c = virtual stage c*pp + r, so the pipeline bubble
+        shrinks ~1/vpp vs plain 1F1B. Every rank enumerates the same
+        global forward/backward slot orders (Megatron-style chunk cycling
+        in pp-sized microbatch groups), which makes the send/recv sequence
+        on each directed channel identical on both ends by construction;
+        sends are isend, recvs block."""
+        ctx = self.ctx
+        pp, v, nm = ctx.pp, self.vpp, self.num_microbatches
+        r = ctx.pp_rank
+        total = nm * v
+        last_vs = pp * v - 1
+        seq = self.spec.seq_length // (ctx.tp if self.sp else 1)
+        act_shape = (self.mbs, seq, self.spec.hidden_size)
+        prev = self._rank_of_stage((r - 1) % pp)
+        nxt = self._rank_of_stage((r + 1) % pp)
+
+        def f_cm(k):
+            return (k // pp) % v, (k // (pp * v)) * pp + k % pp
+
+        def b_cm(k):
+            return v - 1 - (k // pp) % v, (k // (pp * v)) * pp + k % pp
+
+        self.optimizer.zero_grad()
+        pending = []
+        live = {}
+        losses: List[torch.Tensor] = []
+
+        def fwd(k: int) -> None:
+            c, mb = f_cm(k)
+            chunk = self.model_chunks[c]
+            vs = c * pp + r
+            inp = None
+            if vs == 0:
+                x, _ = self.synthetic_batch()
+            else:
+                buf = torch.empty(act_shape, dtype=self.dtype,
+                                  device=ctx.device or "cpu")
+                dist.recv(buf, src=prev)
+                x = buf.requires_grad_(True)
+                inp = x
+            if vs == last_vs:
+                _, labels = self.synthetic_batch()
+                out = chunk(x, labels=labels)
+                losses.append(out.detach())
+            else:
+                out = chunk(x)
+                t = out.detach().contiguous()
+                pending.append((dist.isend(t, dst=nxt), t))
+            aux = None
+            consume = getattr(chunk, "consume_aux_loss", None)
+            if consume is not None:
+                aux = consume()
+                if vs == last_vs:
+                    aux = None  # folded into the loss by the chunk itself
+            live[(c, mb)] = (inp, out, aux)
+
+        def bwd(k: int, is_last_slot: bool) -> None:
+            c, mb = b_cm(k)
+            vs = c * pp + r
+            if self.grad_sync is not None and is_last_slot:
+                self.grad_sync.arm()
+            inp, out, aux = live.pop((c, mb))
+            if vs == last_vs:
+                (out / nm).backward()
+            else:
+                gout = torch.empty(act_shape, dtype=self.dtype,
+                                   device=ctx.device or "cpu")
+                dist.recv(gout, src=nxt)
+                self._backward_stage(out, gout, aux)
+            if vs > 0:
+                t = inp.grad.contiguous()
+                pending.append((dist.isend(t, dst=prev), t))
+
+        warmup = min((pp - r - 1) * 2 + (v - 1) * pp, total)
+        for k in range(warmup):
+            fwd(k)
+        fk, bk = warmup, 0
+        while fk < total:
+            fwd(fk)
+            bwd(bk, bk == total - 1)
+            fk += 1
+            bk += 1
+        while bk < total:
+            bwd(bk, bk == total - 1)
+            bk += 1
+        for work, _ in pending:
+            work.wait()
+
+        self._sync_and_step()
+        if losses:
+            return float(torch.stack(losses).mean())
+        return 0.0
+
     # --- gradient sync + optimizer ----------------------------------------
     def _sync_and_step(self) -> None:
         with self.tracer.span("grad_sync"):
@@ -334,6 +452,8 @@ class PlanRunner:
     def train_step(self) -> float:
         if self.ctx.pp == 1:
             return self._step_no_pipeline()
+        if self.schedule == "interleaved":
+            return self._step_interleaved()
         if self.schedule == "1f1b":
             return self._step_pipeline_1f1b()
         return self._step_pipeline()
@@ -345,7 +465,7 @@ class PlanRunner:
             "model": self.model.state_dict(),
             "optimizer": self.optimizer.state_dict(),
             "plan": {"dp": self.ctx.dp, "tp": self.ctx.tp, "pp": self.ctx.pp,
-                     "mbs": self.mbs, "gbs": self.gbs,
+                     "mbs": self.mbs, "gbs": self.gbs, "vpp": self.vpp,
                      "layer_partition": self.layer_partition},
             "rank": self.ctx.rank,
             # synthetic-data generator state: resume replays the exact
