@@ -75,6 +75,8 @@ def main() -> None:
     p.add_argument("--recompute", action="store_true")
     p.add_argument("--zero1", action="store_true")
     p.add_argument("--sp", action="store_true")
+    p.add_argument("--clip-grad", type=float, default=None,
+                   help="global grad-norm clip (TP/PP-correct)")
     p.add_argument("--layer-partition", default=None)
     p.add_argument("--log-every", type=int, default=10)
     args = p.parse_args()
@@ -86,7 +88,8 @@ def main() -> None:
     runner = PlanRunner(MODEL_SPECS[args.model], ctx, mbs=args.mbs,
                         gbs=args.gbs, layer_partition=lp,
                         schedule=args.schedule, recompute=args.recompute,
-                        zero1=args.zero1, sp=args.sp, vpp=args.vpp)
+                        zero1=args.zero1, sp=args.sp, vpp=args.vpp,
+                        clip_grad=args.clip_grad)
 
     start = 0
     if args.resume and args.checkpoint_dir:

@@ -25,6 +25,7 @@ from metis_amd.models.llama import LlamaModel, LlamaModelSpec
 from metis_amd.models.moe import MoEModel, MoEModelSpec
 from metis_amd.ops import FusedAdamW
 from metis_amd.planner.volume import uniform_layer_split
+from metis_amd.runtime.clip import global_grad_norm, shard_flags
 from metis_amd.runtime.comm import ParallelContext
 from metis_amd.runtime.grad_sync import GradBucketSync
 from metis_amd.runtime.trace import tracer_from_env
@@ -45,6 +46,7 @@ class PlanRunner:
         zero1: bool = False,
         sp: bool = False,
         vpp: int = 1,
+        clip_grad: Optional[float] = None,
     ) -> None:
         assert schedule in ("gpipe", "1f1b", "interleaved")
         self.schedule = schedule
@@ -105,6 +107,17 @@ class PlanRunner:
         shard_group = ctx.dp_group if (zero1 and ctx.dp > 1) else None
         self.optimizer = FusedAdamW(self.model.parameters(), lr=lr,
                                     shard_group=shard_group)
+        self.clip_grad = clip_grad
+        if clip_grad is not None:
+            # classify each optimizer param: True if its elements are a
+            # distinct shard per TP/EP rank (column/row-parallel weights,
+            # expert weights), False if replicated (norms, embeddings,
+            # routers, row-parallel biases) — the global grad norm sums
+            # sharded contributions ACROSS the tp group but replicated
+            # ones once
+            flags = shard_flags(self.model)
+            self._grad_sharded = [flags.get(id(p), False)
+                                  for p in self.optimizer.params]
         self.dtype = dtype
         self.tracer = tracer_from_env(ctx.rank)
         self._check_sync = os.environ.get("METIS_CHECK_SYNC") == "1"
@@ -414,6 +427,13 @@ class PlanRunner:
         return 0.0
 
     # --- gradient sync + optimizer ----------------------------------------
+    def _global_grad_norm(self) -> float:
+        """Global L2 norm of the (DP-synced) flat gradient (runtime.clip)."""
+        flat = self.optimizer.grad_flat
+        pairs = [(flat[off:off + n], sh) for sh, (off, n) in
+                 zip(self._grad_sharded, self.optimizer._slices)]
+        return global_grad_norm(pairs, self.ctx.tp_group, self.ctx.pp_group)
+
     def _sync_and_step(self) -> None:
         with self.tracer.span("grad_sync"):
             if self.grad_sync is not None:
@@ -422,8 +442,13 @@ class PlanRunner:
                 self.grad_sync.finish()
             else:
                 self.optimizer.gather_grads()
+        scale = 1.0
+        if self.clip_grad is not None:
+            norm = self._global_grad_norm()
+            if norm > self.clip_grad:
+                scale = self.clip_grad / (norm + 1e-6)
         with self.tracer.span("optimizer"):
-            self.optimizer.step(pre_gathered=True)
+            self.optimizer.step(pre_gathered=True, grad_scale=scale)
         if self._check_sync:
             self.verify_replicas_synced()
 

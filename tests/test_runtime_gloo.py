@@ -378,3 +378,35 @@ def test_composed_sp_zero1_recompute_dp_tp():
     interactions (SP grad hooks vs bucketed DP sync vs sharded optimizer)
     hold and DP replicas stay bitwise-synced."""
     _run_workers(_composed_worker, world=4, port=29620)
+
+
+def _clip_worker(rank, world, port, out):
+    _env(rank, world, port)
+    from metis_amd.runtime.clip import global_grad_norm, shard_flags
+    from metis_amd.runtime.comm import init_parallel
+
+    ctx = init_parallel(dp=1, tp=2, pp=1)
+    torch.manual_seed(7)
+    full = GPTModel(SPEC, tp=1, dtype=torch.float32)
+    shard = GPTModel(SPEC, tp=2, dtype=torch.float32, tp_group=ctx.tp_group)
+    _shard_from_full(full, shard, rank, 2)
+
+    g = torch.Generator().manual_seed(11)
+    tokens = torch.randint(0, 512, (2, 32), generator=g)
+    labels = torch.roll(tokens, -1, 1)
+    full(tokens, labels=labels).backward()
+    shard(tokens, labels=labels).backward()
+
+    # independent reference: torch's own clip on the UNsharded model
+    ref = torch.nn.utils.clip_grad_norm_(full.parameters(), 1e9)
+    flags = shard_flags(shard)
+    mine = global_grad_norm(
+        [(p.grad, flags.get(id(p), False)) for p in shard.parameters()],
+        tp_group=ctx.tp_group)
+    assert abs(mine - float(ref)) / float(ref) < 1e-5, (mine, float(ref))
+    out.put(("ok", rank))
+    dist.destroy_process_group()
+
+
+def test_global_grad_norm_matches_unsharded():
+    _run_workers(_clip_worker, port=29623)
