@@ -77,6 +77,11 @@ def main() -> None:
     p.add_argument("--sp", action="store_true")
     p.add_argument("--clip-grad", type=float, default=None,
                    help="global grad-norm clip (TP/PP-correct)")
+    p.add_argument("--lr", type=float, default=1e-4)
+    p.add_argument("--lr-warmup", type=int, default=0,
+                   help="linear warmup steps (then cosine decay to "
+                        "--lr-min over --steps)")
+    p.add_argument("--lr-min", type=float, default=0.0)
     p.add_argument("--layer-partition", default=None)
     p.add_argument("--log-every", type=int, default=10)
     args = p.parse_args()
@@ -86,10 +91,16 @@ def main() -> None:
     lp = ([int(x) for x in args.layer_partition.split(",")]
           if args.layer_partition else None)
     runner = PlanRunner(MODEL_SPECS[args.model], ctx, mbs=args.mbs,
-                        gbs=args.gbs, layer_partition=lp,
+                        gbs=args.gbs, layer_partition=lp, lr=args.lr,
                         schedule=args.schedule, recompute=args.recompute,
                         zero1=args.zero1, sp=args.sp, vpp=args.vpp,
                         clip_grad=args.clip_grad)
+    sched = None
+    if args.lr_warmup:
+        from metis_amd.runtime.lr import WarmupCosineLR
+
+        sched = WarmupCosineLR(runner.optimizer, args.lr, args.lr_warmup,
+                               args.steps, args.lr_min)
 
     start = 0
     if args.resume and args.checkpoint_dir:
@@ -102,11 +113,17 @@ def main() -> None:
             if ctx.rank == 0:
                 print(f"resumed from {path} (step {step})")
 
+    t0 = time.time()
+    tokens_per_step = args.gbs * MODEL_SPECS[args.model].seq_length
     for step in range(start, args.steps):
+        if sched is not None:
+            sched.step(step)
         loss = runner.train_step()
         done = step + 1
         if ctx.rank == ctx.world_size - 1 and done % args.log_every == 0:
-            print(f"step {done}: loss {loss:.4f}", flush=True)
+            tps = tokens_per_step * (done - start) / max(time.time() - t0, 1e-9)
+            print(f"step {done}: loss {loss:.4f} lr {runner.optimizer.lr:.2e} "
+                  f"{tps:,.0f} tok/s", flush=True)
         if (args.checkpoint_dir and args.checkpoint_every > 0
                 and done % args.checkpoint_every == 0):
             save_step(runner, args.checkpoint_dir, done)

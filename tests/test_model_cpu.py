@@ -110,3 +110,23 @@ def test_runner_checkpoint_roundtrip(tmp_path):
     assert torch.equal(master_before, runner.optimizer.master)
     assert runner.optimizer.step_count == step_before
     runner.train_step()  # resumes cleanly
+
+
+def test_warmup_cosine_lr():
+    import math
+
+    from metis_amd.ops import FusedAdamW
+    from metis_amd.runtime.lr import WarmupCosineLR
+
+    import torch
+
+    p = torch.nn.Parameter(torch.ones(4))
+    opt = FusedAdamW([p], lr=0.0)
+    sched = WarmupCosineLR(opt, lr_max=1.0, warmup_steps=10, decay_steps=110,
+                           lr_min=0.1)
+    assert abs(sched.step(0) - 0.1) < 1e-9       # first warmup step
+    assert abs(sched.step(9) - 1.0) < 1e-9       # warmup peak
+    mid = sched.step(60)                          # cosine midpoint
+    assert abs(mid - (0.1 + 0.45 * (1 + math.cos(math.pi / 2)))) < 1e-9
+    assert abs(sched.step(200) - 0.1) < 1e-9     # floor after decay
+    assert opt.lr == sched.lr_at(200)
