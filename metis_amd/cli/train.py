@@ -82,6 +82,8 @@ def main() -> None:
                    help="linear warmup steps (then cosine decay to "
                         "--lr-min over --steps)")
     p.add_argument("--lr-min", type=float, default=0.0)
+    p.add_argument("--data", default=None,
+                   help="flat uint16 token file (default: synthetic)")
     p.add_argument("--layer-partition", default=None)
     p.add_argument("--log-every", type=int, default=10)
     args = p.parse_args()
@@ -94,7 +96,8 @@ def main() -> None:
                         gbs=args.gbs, layer_partition=lp, lr=args.lr,
                         schedule=args.schedule, recompute=args.recompute,
                         zero1=args.zero1, sp=args.sp, vpp=args.vpp,
-                        clip_grad=args.clip_grad)
+                        clip_grad=args.clip_grad,
+                        data_path=args.data)
     sched = None
     if args.lr_warmup:
         from metis_amd.runtime.lr import WarmupCosineLR

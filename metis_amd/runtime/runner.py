@@ -47,6 +47,7 @@ class PlanRunner:
         sp: bool = False,
         vpp: int = 1,
         clip_grad: Optional[float] = None,
+        data_path: Optional[str] = None,
     ) -> None:
         assert schedule in ("gpipe", "1f1b", "interleaved")
         self.schedule = schedule
@@ -122,6 +123,16 @@ class PlanRunner:
         self.tracer = tracer_from_env(ctx.rank)
         self._check_sync = os.environ.get("METIS_CHECK_SYNC") == "1"
         self._data_gen = None
+        # real-data path: memory-mapped token file with the same
+        # determinism contract as the synthetic stream (data/dataset.py)
+        self.data_loader = None
+        if data_path is not None:
+            from metis_amd.data import TokenDataset, TokenLoader
+
+            self.data_loader = TokenLoader(
+                TokenDataset(data_path, spec.seq_length), mbs=mbs,
+                dp=ctx.dp, dp_rank=ctx.dp_rank,
+                device=ctx.device or torch.device("cpu"))
         # bucketed, overlapped DP gradient all-reduce (dp > 1 only: the
         # per-parameter Python hooks cost more than the serial gather saves
         # when there is no collective to overlap — measured +40 ms/step)
@@ -147,6 +158,11 @@ class PlanRunner:
         labels = torch.roll(tokens, -1, dims=1)
         return tokens, labels
 
+    def next_batch(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        if self.data_loader is not None:
+            return self.data_loader.next_batch()
+        return self.synthetic_batch()
+
     # --- single-stage step (pp == 1) --------------------------------------
     def _step_no_pipeline(self) -> float:
         losses = []
@@ -154,7 +170,7 @@ class PlanRunner:
         for mb in range(self.num_microbatches):
             if self.grad_sync is not None and mb == self.num_microbatches - 1:
                 self.grad_sync.arm()
-            tokens, labels = self.synthetic_batch()
+            tokens, labels = self.next_batch()
             with self.tracer.span("forward"):
                 loss = self.model(tokens, labels=labels)
             with self.tracer.span("backward"):
@@ -222,7 +238,7 @@ class PlanRunner:
         # forward fill
         for mb in range(nm):
             if ctx.is_first_stage:
-                tokens, labels = self.synthetic_batch()
+                tokens, labels = self.next_batch()
                 x = tokens
                 inputs.append(None)
             else:
@@ -235,7 +251,7 @@ class PlanRunner:
                 else:
                     # labels generated on the last stage (same per-replica
                     # seeded stream as the first stage's tokens)
-                    _, labels = self.synthetic_batch()
+                    _, labels = self.next_batch()
                 out = self.model(x, labels=labels)
                 losses.append(out)
                 outputs.append(out)
@@ -284,12 +300,12 @@ class PlanRunner:
         def fwd(i: int) -> None:
             inp = None
             if ctx.is_first_stage:
-                x, _ = self.synthetic_batch()
+                x, _ = self.next_batch()
             else:
                 x = self._recv_activation(act_shape, prev).requires_grad_(True)
                 inp = x
             if ctx.is_last_stage:
-                _, labels = self.synthetic_batch()
+                _, labels = self.next_batch()
                 out = self.model(x, labels=labels)
                 losses.append(out.detach())
             else:
@@ -366,7 +382,7 @@ class PlanRunner:
             vs = c * pp + r
             inp = None
             if vs == 0:
-                x, _ = self.synthetic_batch()
+                x, _ = self.next_batch()
             else:
                 buf = torch.empty(act_shape, dtype=self.dtype,
                                   device=ctx.device or "cpu")
@@ -374,7 +390,7 @@ class PlanRunner:
                 x = buf.requires_grad_(True)
                 inp = x
             if vs == last_vs:
-                _, labels = self.synthetic_batch()
+                _, labels = self.next_batch()
                 out = chunk(x, labels=labels)
                 losses.append(out.detach())
             else:
@@ -497,6 +513,8 @@ class PlanRunner:
             # token stream a continuous run would have drawn
             "data_gen": (self._data_gen.get_state()
                          if self._data_gen is not None else None),
+            "data_cursor": (self.data_loader.state()
+                            if self.data_loader is not None else None),
         }, path)
 
     def load_checkpoint(self, path: str) -> None:
@@ -510,6 +528,8 @@ class PlanRunner:
             dev = self.ctx.device or torch.device("cpu")
             self._data_gen = torch.Generator(device=dev)
             self._data_gen.set_state(state["data_gen"])
+        if state.get("data_cursor") is not None and self.data_loader is not None:
+            self.data_loader.load_state(int(state["data_cursor"]))
         opt_state = state["optimizer"]
         self.optimizer.load_state_dict({
             "step": opt_state["step"],
