@@ -84,6 +84,10 @@ def main() -> None:
     p.add_argument("--lr-min", type=float, default=0.0)
     p.add_argument("--data", default=None,
                    help="flat uint16 token file (default: synthetic)")
+    p.add_argument("--eval-every", type=int, default=0,
+                   help="eval-loss interval in steps (needs --data; the "
+                        "last 2%% of the file is held out)")
+    p.add_argument("--eval-batches", type=int, default=8)
     p.add_argument("--layer-partition", default=None)
     p.add_argument("--log-every", type=int, default=10)
     args = p.parse_args()
@@ -116,6 +120,26 @@ def main() -> None:
             if ctx.rank == 0:
                 print(f"resumed from {path} (step {step})")
 
+    eval_loader = None
+    if args.eval_every and args.data and ctx.pp == 1:
+        from metis_amd.data import TokenDataset, TokenLoader
+
+        spec = MODEL_SPECS[args.model]
+        eval_loader = TokenLoader(
+            TokenDataset(args.data, spec.seq_length, split=(0.98, 1.0)),
+            mbs=args.mbs, dp=ctx.dp, dp_rank=ctx.dp_rank,
+            device=ctx.device or None)
+
+    def eval_loss() -> float:
+        eval_loader.load_state(0)
+        total = 0.0
+        n = min(args.eval_batches, eval_loader.microbatches_per_epoch)
+        with torch.no_grad():
+            for _ in range(n):
+                tokens, labels = eval_loader.next_batch()
+                total += float(runner.model(tokens, labels=labels))
+        return total / n
+
     t0 = time.time()
     tokens_per_step = args.gbs * MODEL_SPECS[args.model].seq_length
     for step in range(start, args.steps):
@@ -127,6 +151,10 @@ def main() -> None:
             tps = tokens_per_step * (done - start) / max(time.time() - t0, 1e-9)
             print(f"step {done}: loss {loss:.4f} lr {runner.optimizer.lr:.2e} "
                   f"{tps:,.0f} tok/s", flush=True)
+        if eval_loader is not None and done % args.eval_every == 0:
+            el = eval_loss()
+            if ctx.rank == ctx.world_size - 1:
+                print(f"step {done}: eval loss {el:.4f}", flush=True)
         if (args.checkpoint_dir and args.checkpoint_every > 0
                 and done % args.checkpoint_every == 0):
             save_step(runner, args.checkpoint_dir, done)

@@ -24,16 +24,22 @@ import torch
 
 
 class TokenDataset:
-    def __init__(self, path: str, seq_length: int, dtype: str = "uint16"):
+    def __init__(self, path: str, seq_length: int, dtype: str = "uint16",
+                 split=(0.0, 1.0)):
+        """``split``: fraction range of the sample space this dataset
+        covers — e.g. train (0, 0.98) / eval (0.98, 1.0) on one file."""
         self.tokens = np.memmap(path, dtype=np.dtype(dtype), mode="r")
         self.seq_length = seq_length
         # sample i = tokens[i*S : i*S + S + 1] (inputs + shifted labels)
-        self.num_samples = (len(self.tokens) - 1) // seq_length
+        total = (len(self.tokens) - 1) // seq_length
+        self._first = int(total * split[0])
+        self.num_samples = int(total * split[1]) - self._first
         if self.num_samples <= 0:
             raise ValueError(f"{path}: too short for seq_length {seq_length}")
 
     def sample(self, idx: int) -> np.ndarray:
         s = self.seq_length
+        idx += self._first
         return np.asarray(self.tokens[idx * s: idx * s + s + 1])
 
 

@@ -71,3 +71,13 @@ def test_runner_trains_on_file_data(tmp_path):
     l2 = runner.train_step()
     assert l1 > 0 and l2 > 0
     assert runner.data_loader.state() == 4  # 2 steps x 2 microbatches
+
+
+def test_dataset_split_disjoint(tmp_path):
+    p = tmp_path / "toks.bin"
+    _write_tokens(str(p), n=32 * 100 + 1)
+    train = TokenDataset(str(p), 32, split=(0.0, 0.9))
+    ev = TokenDataset(str(p), 32, split=(0.9, 1.0))
+    assert train.num_samples == 90 and ev.num_samples == 10
+    assert np.array_equal(ev.sample(0), np.asarray(
+        TokenDataset(str(p), 32).sample(90)))
