@@ -21,6 +21,9 @@ std::vector<torch::Tensor> qkv_rope_split(
 torch::Tensor qkv_rope_split_bwd(
     torch::Tensor dq, torch::Tensor dk, torch::Tensor dv, long head_dim,
     torch::Tensor cos_t, torch::Tensor sin_t);
+torch::Tensor lt_fp8_matmul(
+    torch::Tensor x, torch::Tensor w,
+    torch::Tensor scale_x, torch::Tensor scale_w);
 std::vector<torch::Tensor> lt_fc1_forward(
     torch::Tensor x, torch::Tensor w, torch::Tensor bias);
 std::vector<torch::Tensor> lt_matmul_dgelu_bgrad(
@@ -94,6 +97,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused QKV relayout + RoPE (neox) forward");
     m.def("qkv_rope_split_bwd", &qkv_rope_split_bwd,
           "fused QKV relayout + RoPE backward gather");
+    m.def("lt_fp8_matmul", &lt_fp8_matmul,
+          "fp8 e4m3 GEMM with per-tensor scales, bf16 out");
     m.def("lt_fc1_forward", &lt_fc1_forward,
           "hipblaslt GEMM with GELU_AUX_BIAS epilogue (fc1 fused)");
     m.def("lt_matmul_dgelu_bgrad", &lt_matmul_dgelu_bgrad,

@@ -231,3 +231,19 @@ def test_ce_row_kernels_match_torch():
     oh[owned] = torch.nn.functional.one_hot(labels[owned], 6400).float()
     ref_d = (soft - oh) * float(scale)
     assert torch.allclose(d.float(), ref_d, atol=5e-3)
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(os.environ.get("METIS_EXPERIMENTAL") != "1",
+                    reason="fp8 GEMM path pending GPU validation")
+def test_fp8_matmul_matches_bf16_within_quant_tolerance():
+    from metis_amd.ops.fp8 import fp8_matmul
+
+    torch.manual_seed(0)
+    x = torch.randn(512, 1024, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(2048, 1024, device="cuda", dtype=torch.bfloat16) * 0.05
+    y = fp8_matmul(x, w)
+    ref = x.float() @ w.float().t()
+    # per-tensor e4m3 quantization error; K=1024 averaging keeps it small
+    err = (y.float() - ref).abs().mean() / ref.abs().mean()
+    assert err < 0.05, float(err)
