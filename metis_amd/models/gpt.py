@@ -331,11 +331,39 @@ class GPTBlock(nn.Module):
         self._lt_mlp = _os.environ.get("METIS_FC1_EPILOGUE") == "1"
 
 
-    def forward(self, x: torch.Tensor, tp_group, sp: bool = False) -> torch.Tensor:
+    def _cached_attention(self, q, k, v):
+        """Decode-path attention over cached K/V: q holds only the new
+        tokens (rightmost positions). q_len == 1 attends everything; a
+        longer new chunk gets a prefix+causal mask."""
+        new, total = q.size(2), k.size(2)
+        if new == 1:
+            return F.scaled_dot_product_attention(q, k, v)
+        mask = torch.ones(new, total, dtype=torch.bool, device=q.device)
+        mask = torch.tril(mask, diagonal=total - new)
+        return F.scaled_dot_product_attention(q, k, v, attn_mask=mask)
+
+    def forward(self, x: torch.Tensor, tp_group, sp: bool = False,
+                cache=None, layer_idx: int = 0) -> torch.Tensor:
         """With ``sp`` the block holds [b, s/tp, h] in the norm/residual
         regions: the f/g operators become seq all-gather / reduce-scatter
         (see the SP collectives above); attention always sees the full
-        sequence."""
+        sequence. With ``cache`` (inference) the block runs incrementally:
+        K/V of new tokens are appended to the cache and attention spans
+        the cached prefix."""
+        if cache is not None:
+            residual = x
+            y = self.ln_attn(x)
+            qkv = self.qkv(y, tp_group)
+            q, k, v = qkv_split_transpose(
+                qkv, self.heads_per_rank, self.heads_per_rank, self.head_dim)
+            k, v = cache.append(layer_idx, k, v)
+            attn = self._cached_attention(q, k, v)
+            x = residual + self.proj(heads_merge(attn), tp_group)
+            residual = x
+            y = self.fc1(self.ln_mlp(x), tp_group)
+            x = residual + self.fc2(F.gelu(y, approximate="tanh"), tp_group)
+            return x
+
         residual = x
         y = self.ln_attn(x)
         # per-rank qkv layout: [q heads | k heads | v heads] blocks
@@ -458,15 +486,19 @@ class GPTModel(nn.Module):
                 p.register_post_accumulate_grad_hook(_sync)
 
     def forward(
-        self, x: torch.Tensor, labels: Optional[torch.Tensor] = None
+        self, x: torch.Tensor, labels: Optional[torch.Tensor] = None,
+        cache=None, pos_offset: int = 0,
     ) -> torch.Tensor:
         """x: token ids [b, s] on the first stage, hidden states elsewhere
         ([b, s/tp, h] between blocks when sp). Returns the loss when this
         stage has the head and labels are given, otherwise the stage's
-        output hidden states."""
+        output hidden states. ``cache``/``pos_offset`` select the
+        incremental KV-cache inference path (runtime.generate)."""
         if self.has_embedding:
             b, s = x.shape
-            if self.sp:
+            if cache is not None:
+                pos = torch.arange(pos_offset, pos_offset + s, device=x.device)
+            elif self.sp:
                 # embed only this rank's sequence slice
                 r = dist.get_rank(self.tp_group)
                 ss = s // self.tp
@@ -477,8 +509,10 @@ class GPTModel(nn.Module):
             x = self.wte(x) + self.wpe(pos)[None, :, :]
 
         use_ckpt = self.recompute and torch.is_grad_enabled()
-        for block in self.blocks:
-            if use_ckpt:
+        for i, block in enumerate(self.blocks):
+            if cache is not None:
+                x = block(x, self.tp_group, cache=cache, layer_idx=i)
+            elif use_ckpt:
                 x = torch.utils.checkpoint.checkpoint(
                     block, x, self.tp_group, self.sp, use_reentrant=False)
             else:

@@ -91,8 +91,39 @@ class LlamaBlock(nn.Module):
         self._fused_qkv_rope = _os.environ.get("METIS_QKV_ROPE") == "1"
 
 
-    def forward(self, x: torch.Tensor, tp_group) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, tp_group, cache=None,
+                layer_idx: int = 0, pos_offset: int = 0) -> torch.Tensor:
         hq, hkv, d = self.heads_per_rank, self.kv_heads_per_rank, self.head_dim
+
+        if cache is not None:
+            import torch.nn.functional as F
+
+            residual = x
+            y = self.norm_attn(x)
+            qkv = self.qkv(y, tp_group)
+            q, k, v = qkv_split_transpose(qkv, hq, hkv, d)
+            q = apply_rope(q, self.rope_base, pos_offset=pos_offset)
+            k = apply_rope(k, self.rope_base, pos_offset=pos_offset)
+            k, v = cache.append(layer_idx, k, v)
+            if hkv != hq:  # GQA: expand kv heads for SDPA
+                rep = hq // hkv
+                k = k.repeat_interleave(rep, dim=1)
+                v = v.repeat_interleave(rep, dim=1)
+            new, total = q.size(2), k.size(2)
+            if new == 1:
+                attn = F.scaled_dot_product_attention(q, k, v)
+            else:
+                mask = torch.tril(
+                    torch.ones(new, total, dtype=torch.bool, device=q.device),
+                    diagonal=total - new)
+                attn = F.scaled_dot_product_attention(q, k, v, attn_mask=mask)
+            x = residual + self.proj(heads_merge(attn), tp_group)
+            residual = x
+            gate_up = self.gate_up(self.norm_mlp(x), tp_group)
+            gate, up = gate_up.split([self.ffn_per_rank, self.ffn_per_rank],
+                                     dim=-1)
+            return residual + self.down(
+                swiglu(gate.contiguous(), up.contiguous()), tp_group)
 
         residual = x
         y = self.norm_attn(x)
@@ -152,12 +183,15 @@ class LlamaModel(nn.Module):
 
         self.recompute = False  # see GPTModel.recompute
 
-    def forward(self, x, labels=None):
+    def forward(self, x, labels=None, cache=None, pos_offset: int = 0):
         if self.has_embedding:
             x = self.wte(x)
         use_ckpt = self.recompute and torch.is_grad_enabled()
-        for block in self.blocks:
-            if use_ckpt:
+        for i, block in enumerate(self.blocks):
+            if cache is not None:
+                x = block(x, self.tp_group, cache=cache, layer_idx=i,
+                          pos_offset=pos_offset)
+            elif use_ckpt:
                 x = torch.utils.checkpoint.checkpoint(
                     block, x, self.tp_group, use_reentrant=False)
             else:

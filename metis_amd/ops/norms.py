@@ -86,12 +86,16 @@ def _rope_ref(x: torch.Tensor, cos_t: torch.Tensor, sin_t: torch.Tensor) -> torc
     return torch.cat((x1 * c - x2 * sn, x2 * c + x1 * sn), dim=-1)
 
 
-def apply_rope(x: torch.Tensor, base: float = 500000.0) -> torch.Tensor:
-    """x [B, H, S, D] -> rotated x (llama half-rotation convention)."""
+def apply_rope(x: torch.Tensor, base: float = 500000.0,
+               pos_offset: int = 0) -> torch.Tensor:
+    """x [B, H, S, D] -> rotated x (llama half-rotation convention);
+    ``pos_offset`` shifts the absolute positions (KV-cache decode)."""
     s, d = x.size(2), x.size(3)
-    cos_t, sin_t = rope_tables(s, d, base, x.device)
+    cos_t, sin_t = rope_tables(pos_offset + s, d, base, x.device)
+    cos_t, sin_t = cos_t[pos_offset:], sin_t[pos_offset:]
     if x.is_cuda and x.dtype == torch.bfloat16:
-        return _RoPEFn.apply(x.contiguous(), cos_t, sin_t)
+        return _RoPEFn.apply(x.contiguous(), cos_t.contiguous(),
+                             sin_t.contiguous())
     return _rope_ref(x, cos_t, sin_t)
 
 

@@ -410,3 +410,27 @@ def _clip_worker(rank, world, port, out):
 
 def test_global_grad_norm_matches_unsharded():
     _run_workers(_clip_worker, port=29623)
+
+
+def _generate_tp_worker(rank, world, port, out):
+    _env(rank, world, port)
+    from metis_amd.runtime.comm import init_parallel
+    from metis_amd.runtime.generate import generate
+
+    ctx = init_parallel(dp=1, tp=2, pp=1)
+    torch.manual_seed(7)
+    full = GPTModel(SPEC, tp=1, dtype=torch.float32)
+    shard = GPTModel(SPEC, tp=2, dtype=torch.float32, tp_group=ctx.tp_group)
+    _shard_from_full(full, shard, rank, 2)
+
+    g = torch.Generator().manual_seed(11)
+    prompt = torch.randint(0, 512, (1, 8), generator=g)
+    ref = generate(full, prompt, 6, temperature=0.0)
+    tp = generate(shard, prompt, 6, temperature=0.0, tp_group=ctx.tp_group)
+    assert torch.equal(ref, tp), (ref, tp)
+    out.put(("ok", rank))
+    dist.destroy_process_group()
+
+
+def test_generate_tensor_parallel_matches_single():
+    _run_workers(_generate_tp_worker, port=29624)
