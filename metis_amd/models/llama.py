@@ -16,6 +16,8 @@ import torch.nn as nn
 from metis_amd.models.gpt import (
     ColumnParallelLinear,
     RowParallelLinear,
+    _GatherSeq,
+    _ReduceScatterSeq,
     vocab_parallel_ce,
 )
 from metis_amd.ops.attention import flash_attention
@@ -91,8 +93,9 @@ class LlamaBlock(nn.Module):
         self._fused_qkv_rope = _os.environ.get("METIS_QKV_ROPE") == "1"
 
 
-    def forward(self, x: torch.Tensor, tp_group, cache=None,
-                layer_idx: int = 0, pos_offset: int = 0) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, tp_group, sp: bool = False,
+                cache=None, layer_idx: int = 0,
+                pos_offset: int = 0) -> torch.Tensor:
         hq, hkv, d = self.heads_per_rank, self.kv_heads_per_rank, self.head_dim
 
         if cache is not None:
@@ -125,9 +128,15 @@ class LlamaBlock(nn.Module):
             return residual + self.down(
                 swiglu(gate.contiguous(), up.contiguous()), tp_group)
 
+        import torch.nn.functional as F
+
         residual = x
         y = self.norm_attn(x)
-        qkv = self.qkv(y, tp_group)
+        if sp:  # seq all-gather replaces the f operator (see gpt.py SP)
+            y = _GatherSeq.apply(y, tp_group)
+            qkv = F.linear(y, self.qkv.weight, self.qkv.bias)
+        else:
+            qkv = self.qkv(y, tp_group)
         if self._fused_qkv_rope:
             q, k, v = qkv_rope_split(qkv, hq, hkv, d, self.rope_base)
         else:
@@ -135,14 +144,28 @@ class LlamaBlock(nn.Module):
             q = apply_rope(q, self.rope_base)
             k = apply_rope(k, self.rope_base)
         attn = flash_attention(q, k, v, causal=True)
-        x = residual + self.proj(heads_merge(attn), tp_group)
+        if sp:
+            part = F.linear(heads_merge(attn), self.proj.weight)
+            part = _ReduceScatterSeq.apply(part, tp_group)
+            x = residual + part + self.proj.bias
+        else:
+            x = residual + self.proj(heads_merge(attn), tp_group)
 
         residual = x
         y = self.norm_mlp(x)
-        gate_up = self.gate_up(y, tp_group)
+        if sp:
+            y = _GatherSeq.apply(y, tp_group)
+            gate_up = F.linear(y, self.gate_up.weight, self.gate_up.bias)
+        else:
+            gate_up = self.gate_up(y, tp_group)
         gate, up = gate_up.split([self.ffn_per_rank, self.ffn_per_rank], dim=-1)
-        x = residual + self.down(swiglu(gate.contiguous(), up.contiguous()),
-                                 tp_group)
+        act = swiglu(gate.contiguous(), up.contiguous())
+        if sp:
+            part = _ReduceScatterSeq.apply(F.linear(act, self.down.weight),
+                                           tp_group)
+            x = residual + part + self.down.bias
+        else:
+            x = residual + self.down(act, tp_group)
         return x
 
 
@@ -156,11 +179,15 @@ class LlamaModel(nn.Module):
         dtype: torch.dtype = torch.bfloat16,
         layer_range: Optional[tuple] = None,
         tp_group=None,
+        sp: bool = False,
     ):
         super().__init__()
         self.spec = spec
         self.tp = tp
         self.tp_group = tp_group
+        self.sp = sp and tp > 1
+        if self.sp:
+            assert spec.seq_length % tp == 0, "sp needs seq % tp == 0"
         total = spec.profile_num_layers
         start, end = layer_range if layer_range is not None else (0, total)
         self.has_embedding = start == 0
@@ -183,8 +210,31 @@ class LlamaModel(nn.Module):
 
         self.recompute = False  # see GPTModel.recompute
 
+        # SP grad-sum hooks for replicated params that see only a seq
+        # slice (see GPTModel; llama adds rmsnorm weights + rp biases)
+        if self.sp and tp_group is not None:
+            hooked = []
+            if self.has_embedding:
+                hooked.append(self.wte.weight)
+            for blk in self.blocks:
+                hooked += [blk.norm_attn.weight, blk.norm_mlp.weight,
+                           blk.proj.bias, blk.down.bias]
+            if self.has_head:
+                hooked.append(self.norm_final.weight)
+
+            def _sync(p):
+                if p.grad is not None:
+                    dist.all_reduce(p.grad, group=tp_group)
+
+            for p in hooked:
+                p.register_post_accumulate_grad_hook(_sync)
+
     def forward(self, x, labels=None, cache=None, pos_offset: int = 0):
         if self.has_embedding:
+            if self.sp and cache is None:
+                r = dist.get_rank(self.tp_group)
+                ss = x.size(1) // self.tp
+                x = x[:, r * ss:(r + 1) * ss]
             x = self.wte(x)
         use_ckpt = self.recompute and torch.is_grad_enabled()
         for i, block in enumerate(self.blocks):
@@ -193,12 +243,19 @@ class LlamaModel(nn.Module):
                           pos_offset=pos_offset)
             elif use_ckpt:
                 x = torch.utils.checkpoint.checkpoint(
-                    block, x, self.tp_group, use_reentrant=False)
+                    block, x, self.tp_group, self.sp, use_reentrant=False)
             else:
-                x = block(x, self.tp_group)
+                x = block(x, self.tp_group, self.sp)
         if self.has_head:
-            x = self.norm_final(x)
-            logits = self.head(x, self.tp_group)
+            if self.sp and cache is None:
+                import torch.nn.functional as F
+
+                x = _GatherSeq.apply(x, self.tp_group)
+                x = self.norm_final(x)
+                logits = F.linear(x, self.head.weight, self.head.bias)
+            else:
+                x = self.norm_final(x)
+                logits = self.head(x, self.tp_group)
             if labels is not None:
                 labels = labels.reshape(-1)
                 if self.tp_group is not None and dist.get_world_size(self.tp_group) > 1:

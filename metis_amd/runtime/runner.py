@@ -84,10 +84,12 @@ class PlanRunner:
             model_cls = LlamaModel
         else:
             model_cls = GPTModel
-        self.sp = bool(sp) and ctx.tp > 1 and model_cls is GPTModel
-        extra = {"sp": self.sp} if model_cls is GPTModel else {}
+        sp_capable = model_cls in (GPTModel, LlamaModel)
+        self.sp = bool(sp) and ctx.tp > 1 and sp_capable
+        extra = {"sp": self.sp} if sp_capable else {}
         assert not (sp and not self.sp and ctx.tp > 1), (
-            "sequence parallelism is implemented for the GPT family")
+            "sequence parallelism is implemented for the GPT and Llama "
+            "families")
 
         # one model slice per virtual chunk: chunk c covers virtual stage
         # c*pp + pp_rank (vpp == 1 -> the ordinary single slice)

@@ -434,3 +434,66 @@ def _generate_tp_worker(rank, world, port, out):
 
 def test_generate_tensor_parallel_matches_single():
     _run_workers(_generate_tp_worker, port=29624)
+
+
+def _sp_llama_worker(rank, world, port, out):
+    _env(rank, world, port)
+    from metis_amd.models.llama import LlamaModel, LLAMA_SPECS
+    from metis_amd.runtime.comm import init_parallel
+
+    spec = LLAMA_SPECS["llama-tiny"]
+    ctx = init_parallel(dp=1, tp=2, pp=1)
+    torch.manual_seed(7)
+    full = LlamaModel(spec, tp=1, dtype=torch.float32)
+    shard = LlamaModel(spec, tp=2, dtype=torch.float32,
+                       tp_group=ctx.tp_group, sp=True)
+    # copy weights: llama sharding (qkv gqa blocks, gate_up halves)
+    with torch.no_grad():
+        shard.wte.weight.copy_(full.wte.weight)
+        for fb, sb in zip(full.blocks, shard.blocks):
+            sb.norm_attn.weight.copy_(fb.norm_attn.weight)
+            sb.norm_mlp.weight.copy_(fb.norm_mlp.weight)
+            d = spec.head_dim
+            nq, nkv = spec.num_heads, spec.num_kv_heads
+            qh, kh = nq // 2, nkv // 2
+            rows = torch.cat([
+                torch.arange(rank * qh * d, (rank + 1) * qh * d),
+                nq * d + torch.arange(rank * kh * d, (rank + 1) * kh * d),
+                (nq + nkv) * d + torch.arange(rank * kh * d, (rank + 1) * kh * d),
+            ])
+            sb.qkv.weight.copy_(fb.qkv.weight[rows])
+            sb.qkv.bias.copy_(fb.qkv.bias[rows])
+            ipr = sb.proj.in_per_rank
+            sb.proj.weight.copy_(fb.proj.weight[:, rank * ipr:(rank + 1) * ipr])
+            sb.proj.bias.copy_(fb.proj.bias)
+            f = spec.ffn_hidden_size
+            fp = f // 2
+            gu_rows = torch.cat([torch.arange(rank * fp, (rank + 1) * fp),
+                                 f + torch.arange(rank * fp, (rank + 1) * fp)])
+            sb.gate_up.weight.copy_(fb.gate_up.weight[gu_rows])
+            sb.gate_up.bias.copy_(fb.gate_up.bias[gu_rows])
+            sb.down.weight.copy_(fb.down.weight[:, rank * fp:(rank + 1) * fp])
+            sb.down.bias.copy_(fb.down.bias)
+        shard.norm_final.weight.copy_(full.norm_final.weight)
+        opr = shard.head.out_per_rank
+        shard.head.weight.copy_(full.head.weight[rank * opr:(rank + 1) * opr])
+        shard.head.bias.copy_(full.head.bias[rank * opr:(rank + 1) * opr])
+
+    g = torch.Generator().manual_seed(11)
+    tokens = torch.randint(0, spec.vocab_size, (2, 16), generator=g)
+    labels = torch.roll(tokens, -1, 1)
+    ref = full(tokens, labels=labels)
+    sp = shard(tokens, labels=labels)
+    assert torch.allclose(ref, sp, atol=1e-4), (ref, sp)
+    ref.backward()
+    sp.backward()
+    assert torch.allclose(full.blocks[0].norm_attn.weight.grad,
+                          shard.blocks[0].norm_attn.weight.grad, atol=1e-4)
+    assert torch.allclose(full.wte.weight.grad, shard.wte.weight.grad,
+                          atol=1e-4)
+    out.put(("ok", rank))
+    dist.destroy_process_group()
+
+
+def test_sequence_parallel_llama_matches_single():
+    _run_workers(_sp_llama_worker, port=29625)
