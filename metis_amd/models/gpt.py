@@ -24,7 +24,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from metis_amd.ops import LayerNorm
-from metis_amd.ops.attention import flash_attention
+from metis_amd.ops.attention import decode_attention, flash_attention
 from metis_amd.ops.cross_entropy import cross_entropy
 from metis_amd.ops.mlp import fused_mlp
 from metis_amd.ops.relayout import heads_merge, qkv_split_transpose
@@ -337,7 +337,7 @@ class GPTBlock(nn.Module):
         longer new chunk gets a prefix+causal mask."""
         new, total = q.size(2), k.size(2)
         if new == 1:
-            return F.scaled_dot_product_attention(q, k, v)
+            return decode_attention(q, k, v)
         mask = torch.ones(new, total, dtype=torch.bool, device=q.device)
         mask = torch.tril(mask, diagonal=total - new)
         return F.scaled_dot_product_attention(q, k, v, attn_mask=mask)

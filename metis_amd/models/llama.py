@@ -20,7 +20,7 @@ from metis_amd.models.gpt import (
     _ReduceScatterSeq,
     vocab_parallel_ce,
 )
-from metis_amd.ops.attention import flash_attention
+from metis_amd.ops.attention import decode_attention, flash_attention
 from metis_amd.ops.cross_entropy import cross_entropy
 from metis_amd.ops.norms import RMSNorm, apply_rope, swiglu
 from metis_amd.ops.relayout import (heads_merge, qkv_rope_split,
@@ -108,14 +108,14 @@ class LlamaBlock(nn.Module):
             q = apply_rope(q, self.rope_base, pos_offset=pos_offset)
             k = apply_rope(k, self.rope_base, pos_offset=pos_offset)
             k, v = cache.append(layer_idx, k, v)
-            if hkv != hq:  # GQA: expand kv heads for SDPA
-                rep = hq // hkv
-                k = k.repeat_interleave(rep, dim=1)
-                v = v.repeat_interleave(rep, dim=1)
             new, total = q.size(2), k.size(2)
             if new == 1:
-                attn = F.scaled_dot_product_attention(q, k, v)
+                attn = decode_attention(q, k, v)  # GQA mapped inside
             else:
+                if hkv != hq:  # GQA: expand kv heads for SDPA
+                    rep = hq // hkv
+                    k = k.repeat_interleave(rep, dim=1)
+                    v = v.repeat_interleave(rep, dim=1)
                 mask = torch.tril(
                     torch.ones(new, total, dtype=torch.bool, device=q.device),
                     diagonal=total - new)

@@ -73,3 +73,23 @@ def flash_attention(
     return F.scaled_dot_product_attention(
         q, k, v, is_causal=causal, scale=scale, enable_gqa=q.size(1) != k.size(1)
     )
+
+
+def decode_attention(q: torch.Tensor, k: torch.Tensor,
+                     v: torch.Tensor) -> torch.Tensor:
+    """Single-query attention over cached K/V (serving decode step):
+    q [B, H, 1, D] vs k/v [B, Hkv, S, D], GQA mapped in-kernel. The
+    gfx950 kernel (decode.hip) runs under METIS_DECODE_KERNEL=1 until
+    GPU-validated; otherwise SDPA over the (expanded) cache."""
+    import os
+
+    if (q.is_cuda and q.dtype == torch.bfloat16 and q.size(2) == 1
+            and q.size(-1) in (64, 80, 96, 128)
+            and os.environ.get("METIS_DECODE_KERNEL") == "1"):
+        ext = _ops.require_extension()
+        return ext.attn_decode(q, k, v, 1.0 / math.sqrt(q.size(-1)))
+    if k.size(1) != q.size(1):
+        rep = q.size(1) // k.size(1)
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    return F.scaled_dot_product_attention(q, k, v)

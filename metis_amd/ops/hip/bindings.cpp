@@ -13,6 +13,8 @@ void adamw_step(
     torch::Tensor m, torch::Tensor v,
     double lr, double beta1, double beta2, double eps,
     double weight_decay, long step, double grad_scale);
+torch::Tensor attn_decode(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v, double scale);
 torch::Tensor ce_row_max(torch::Tensor logits);
 torch::Tensor ce_row_sumexp(torch::Tensor logits, torch::Tensor m);
 std::vector<torch::Tensor> qkv_rope_split(
@@ -90,6 +92,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "dq/dk/dv -> fused dqkv layout");
     m.def("heads_merge", &heads_merge, "[B,H,S,D] -> [B,S,H*D]");
     m.def("heads_unmerge", &heads_unmerge, "[B,S,H*D] -> [B,H,S,D]");
+    m.def("attn_decode", &attn_decode,
+          "single-query attention over cached K/V (GQA, online softmax)");
     m.def("ce_row_max", &ce_row_max, "per-row max of bf16 logits (fp32)");
     m.def("ce_row_sumexp", &ce_row_sumexp,
           "per-row sum(exp(x - m)) of bf16 logits (fp32)");

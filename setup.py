@@ -30,6 +30,7 @@ sources = [
     os.path.join(HIP_DIR, "cross_entropy.hip"),
     os.path.join(HIP_DIR, "relayout.hip"),
     os.path.join(HIP_DIR, "qkv_rope.hip"),
+    os.path.join(HIP_DIR, "decode.hip"),
     os.path.join(HIP_DIR, "lt_gemm.cpp"),
 ]
 sources = [s for s in sources if os.path.exists(s)]

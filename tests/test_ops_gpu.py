@@ -247,3 +247,25 @@ def test_fp8_matmul_matches_bf16_within_quant_tolerance():
     # per-tensor e4m3 quantization error; K=1024 averaging keeps it small
     err = (y.float() - ref).abs().mean() / ref.abs().mean()
     assert err < 0.05, float(err)
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(os.environ.get("METIS_EXPERIMENTAL") != "1",
+                    reason="decode attention kernel pending GPU validation")
+def test_attn_decode_matches_sdpa():
+    import math
+
+    import metis_amd._hip_ops as ext
+
+    torch.manual_seed(0)
+    for D, hkv in ((64, 8), (80, 8), (128, 2)):
+        B, H, S = 4, 8, 777
+        q = torch.randn(B, H, 1, D, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn(B, hkv, S, D, device="cuda", dtype=torch.bfloat16)
+        v = torch.randn(B, hkv, S, D, device="cuda", dtype=torch.bfloat16)
+        out = ext.attn_decode(q, k, v, 1.0 / math.sqrt(D))
+        rep = H // hkv
+        ref = torch.nn.functional.scaled_dot_product_attention(
+            q.float(), k.float().repeat_interleave(rep, dim=1),
+            v.float().repeat_interleave(rep, dim=1))
+        assert torch.allclose(out.float(), ref, atol=2e-2), (D, hkv)
