@@ -25,7 +25,7 @@ from metis_amd.models.gpt import GPTModel, MODEL_SPECS as _GPT_SPECS
 from metis_amd.models.llama import LlamaModel, LlamaModelSpec, LLAMA_SPECS
 
 MODEL_SPECS = {**_GPT_SPECS, **LLAMA_SPECS}
-from metis_amd.runtime.generate import generate  # noqa: E402
+from metis_amd.runtime.generate import generate, generate_ragged  # noqa: E402
 
 
 def load_model(model_name: str, checkpoint: str = None,
@@ -70,12 +70,19 @@ def build_app(model, spec, device=None):
 
     @app.post("/generate")
     def gen(req: GenRequest):
+        g = torch.Generator().manual_seed(req.seed) if req.seed else None
+        lens = {len(p) for p in req.tokens}
+        budget = spec.seq_length - max(lens)
+        n = max(0, min(req.max_new_tokens, budget))
+        if len(lens) > 1:   # ragged batch: padded-cache batched decode
+            out = generate_ragged(model, req.tokens, n,
+                                  temperature=req.temperature,
+                                  top_k=req.top_k, generator=g,
+                                  device=device)
+            return {"tokens": out}
         toks = torch.tensor(req.tokens, dtype=torch.long)
         if device is not None:
             toks = toks.to(device)
-        budget = spec.seq_length - toks.size(1)
-        n = max(0, min(req.max_new_tokens, budget))
-        g = torch.Generator().manual_seed(req.seed) if req.seed else None
         out = generate(model, toks, n, temperature=req.temperature,
                        top_k=req.top_k, generator=g)
         return {"tokens": out.cpu().tolist()}

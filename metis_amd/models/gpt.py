@@ -331,11 +331,14 @@ class GPTBlock(nn.Module):
         self._lt_mlp = _os.environ.get("METIS_FC1_EPILOGUE") == "1"
 
 
-    def _cached_attention(self, q, k, v):
+    def _cached_attention(self, q, k, v, mask=None):
         """Decode-path attention over cached K/V: q holds only the new
-        tokens (rightmost positions). q_len == 1 attends everything; a
-        longer new chunk gets a prefix+causal mask."""
+        tokens (rightmost positions). q_len == 1 attends everything (or
+        the ragged cache's padding mask); a longer new chunk gets a
+        prefix+causal mask."""
         new, total = q.size(2), k.size(2)
+        if mask is not None:
+            return F.scaled_dot_product_attention(q, k, v, attn_mask=mask)
         if new == 1:
             return decode_attention(q, k, v)
         mask = torch.ones(new, total, dtype=torch.bool, device=q.device)
@@ -357,7 +360,9 @@ class GPTBlock(nn.Module):
             q, k, v = qkv_split_transpose(
                 qkv, self.heads_per_rank, self.heads_per_rank, self.head_dim)
             k, v = cache.append(layer_idx, k, v)
-            attn = self._cached_attention(q, k, v)
+            am = getattr(cache, "attention_mask", None)
+            mask = am(k.size(2), q.device) if am is not None else None
+            attn = self._cached_attention(q, k, v, mask)
             x = residual + self.proj(heads_merge(attn), tp_group)
             residual = x
             y = self.fc1(self.ln_mlp(x), tp_group)
@@ -497,7 +502,12 @@ class GPTModel(nn.Module):
         if self.has_embedding:
             b, s = x.shape
             if cache is not None:
-                pos = torch.arange(pos_offset, pos_offset + s, device=x.device)
+                if torch.is_tensor(pos_offset):   # ragged: per-row positions
+                    pos = (pos_offset.to(x.device)[:, None]
+                           + torch.arange(s, device=x.device)[None])
+                else:
+                    pos = torch.arange(pos_offset, pos_offset + s,
+                                       device=x.device)
             elif self.sp:
                 # embed only this rank's sequence slice
                 r = dist.get_rank(self.tp_group)
@@ -506,7 +516,8 @@ class GPTModel(nn.Module):
                 pos = torch.arange(r * ss, (r + 1) * ss, device=x.device)
             else:
                 pos = torch.arange(s, device=x.device)
-            x = self.wte(x) + self.wpe(pos)[None, :, :]
+            wpe = self.wpe(pos)
+            x = self.wte(x) + (wpe if wpe.dim() == 3 else wpe[None, :, :])
 
         use_ckpt = self.recompute and torch.is_grad_enabled()
         for i, block in enumerate(self.blocks):

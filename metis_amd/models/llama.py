@@ -22,7 +22,8 @@ from metis_amd.models.gpt import (
 )
 from metis_amd.ops.attention import decode_attention, flash_attention
 from metis_amd.ops.cross_entropy import cross_entropy
-from metis_amd.ops.norms import RMSNorm, apply_rope, swiglu
+from metis_amd.ops.norms import (RMSNorm, apply_rope,
+                                 apply_rope_rows, swiglu)
 from metis_amd.ops.relayout import (heads_merge, qkv_rope_split,
                                     qkv_split_transpose)
 
@@ -105,11 +106,22 @@ class LlamaBlock(nn.Module):
             y = self.norm_attn(x)
             qkv = self.qkv(y, tp_group)
             q, k, v = qkv_split_transpose(qkv, hq, hkv, d)
-            q = apply_rope(q, self.rope_base, pos_offset=pos_offset)
-            k = apply_rope(k, self.rope_base, pos_offset=pos_offset)
+            if torch.is_tensor(pos_offset):       # ragged per-row positions
+                q = apply_rope_rows(q, self.rope_base, pos_offset)
+                k = apply_rope_rows(k, self.rope_base, pos_offset)
+            else:
+                q = apply_rope(q, self.rope_base, pos_offset=pos_offset)
+                k = apply_rope(k, self.rope_base, pos_offset=pos_offset)
             k, v = cache.append(layer_idx, k, v)
+            am = getattr(cache, "attention_mask", None)
+            rag_mask = am(k.size(2), q.device) if am is not None else None
             new, total = q.size(2), k.size(2)
-            if new == 1:
+            if rag_mask is not None:
+                rep = hq // hkv
+                attn = F.scaled_dot_product_attention(
+                    q, k.repeat_interleave(rep, dim=1),
+                    v.repeat_interleave(rep, dim=1), attn_mask=rag_mask)
+            elif new == 1:
                 attn = decode_attention(q, k, v)  # GQA mapped inside
             else:
                 if hkv != hq:  # GQA: expand kv heads for SDPA

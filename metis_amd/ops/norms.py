@@ -120,3 +120,15 @@ def swiglu(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     if a.is_cuda and a.dtype == torch.bfloat16:
         return _SwiGLUFn.apply(a, b)
     return torch.nn.functional.silu(a) * b
+
+
+def apply_rope_rows(x: torch.Tensor, base: float,
+                    positions: torch.Tensor) -> torch.Tensor:
+    """Decode-path RoPE with a PER-ROW position: x [b, h, 1, d],
+    positions [b] (ragged batched decoding)."""
+    d = x.size(-1)
+    cos_t, sin_t = rope_tables(int(positions.max()) + 1, d, base, x.device)
+    c = cos_t[positions].view(-1, 1, 1, d // 2).to(x.dtype)
+    sn = sin_t[positions].view(-1, 1, 1, d // 2).to(x.dtype)
+    x1, x2 = x[..., : d // 2], x[..., d // 2:]
+    return torch.cat((x1 * c - x2 * sn, x2 * c + x1 * sn), dim=-1)

@@ -64,3 +64,29 @@ def test_generate_topk_sampling_seeded():
     s2 = generate(model, prompt, 5, temperature=0.8, top_k=20,
                   generator=torch.Generator().manual_seed(1))
     assert torch.equal(s1, s2)
+
+
+def test_generate_ragged_matches_per_sequence():
+    """Batched ragged decoding (different prompt lengths, padded caches,
+    per-row positions) must reproduce per-sequence greedy decoding
+    exactly — GPT and Llama."""
+    from metis_amd.models.llama import LlamaModel, LLAMA_SPECS
+    from metis_amd.runtime.generate import generate_ragged
+
+    g = torch.Generator().manual_seed(11)
+    prompts = [torch.randint(0, 500, (n,), generator=g).tolist()
+               for n in (5, 9, 3)]
+
+    for build in (
+        lambda: GPTModel(GPT_SPEC, dtype=torch.float32),
+        lambda: LlamaModel(LLAMA_SPECS["llama-tiny"], dtype=torch.float32),
+    ):
+        torch.manual_seed(0)
+        model = build()
+        model.eval()
+        ref = []
+        for p in prompts:
+            toks = torch.tensor([p], dtype=torch.long)
+            ref.append(generate(model, toks, 6, temperature=0.0)[0].tolist())
+        rag = generate_ragged(model, prompts, 6, temperature=0.0)
+        assert rag == ref

@@ -60,3 +60,20 @@ def test_checkpoint_to_serve_roundtrip(tmp_path):
         "tokens": [[1, 2, 3, 4]], "max_new_tokens": 4})
     assert r.status_code == 200
     assert len(r.json()["tokens"][0]) == 8
+
+
+def test_ragged_batch_endpoint():
+    torch.manual_seed(0)
+    model = GPTModel(SPEC, dtype=torch.float32)
+    model.eval()
+    client = _client(model)
+    r = client.post("/generate", json={
+        "tokens": [[1, 2, 3, 4, 5], [9, 8], [7, 6, 5]],
+        "max_new_tokens": 4, "temperature": 0.0})
+    assert r.status_code == 200
+    rows = r.json()["tokens"]
+    assert [len(x) for x in rows] == [9, 6, 7]
+    # each row matches its single-prompt greedy continuation
+    single = client.post("/generate", json={
+        "tokens": [[9, 8]], "max_new_tokens": 4, "temperature": 0.0})
+    assert rows[1] == single.json()["tokens"][0]
