@@ -56,6 +56,24 @@ def compute_balancer(args):
     return out
 
 
+def data_balancer(args):
+    """Fuzz DataLoadBalancer: cases carry synthetic per-type layer-compute
+    totals; build the minimal profile_data dict the reference reads."""
+    from model.load_balancer import DataLoadBalancer
+
+    out = []
+    for device_types, dp_deg, tp_deg, bs, totals in args["cases"]:
+        profile_data = {
+            f"DeviceType.{t}": {
+                f"tp{tp_deg}_bs1": {"time": {"layer-computes": [v]}}
+            }
+            for t, v in totals.items()
+        }
+        bal = DataLoadBalancer(profile_data, None)
+        out.append(bal.partition_data(device_types, (dp_deg, tp_deg), bs))
+    return out
+
+
 def homo_costs(args):
     from gpu_cluster import GPUCluster
     from data_loader import ProfileDataLoader
@@ -86,6 +104,7 @@ def homo_costs(args):
 
 
 MODES = {
+    "data_balancer": data_balancer,
     "device_groups": device_groups,
     "uniform_plans": uniform_plans,
     "compute_balancer": compute_balancer,

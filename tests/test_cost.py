@@ -106,3 +106,33 @@ def test_hetero_cost_runs(store, homo_cluster):
     )
     cost1 = est.get_cost(plan1, [(8, 1)], [0, 10], rank_map)
     assert cost1 > 0
+
+
+def test_json_out_cli(tmp_path, monkeypatch):
+    """--json_out structured plan dump on the hetero CLI."""
+    import json as _json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = tmp_path / "plans.json"
+    cmd = [sys.executable, "cost_het_cluster.py",
+           "--model_name", "GPT", "--num_layers", "10", "--gbs", "16",
+           "--hidden_size", "4096", "--sequence_length", "1024",
+           "--vocab_size", "51200",
+           "--hostfile_path", "tests/data/profiles_synth_hostfile"
+           if os.path.exists(os.path.join(repo, "tests/data/profiles_synth_hostfile"))
+           else "tests/data/mi355x_single_node/hostfile",
+           "--clusterfile_path", "tests/data/mi355x_single_node/clusterfile.json",
+           "--profile_data_path", "tests/data/profiles_synth",
+           "--max_profiled_tp_degree", "4", "--max_profiled_batch_size", "4",
+           "--top_k", "5", "--json_out", str(out)]
+    r = subprocess.run(cmd, cwd=repo, capture_output=True, text=True,
+                       timeout=300)
+    assert r.returncode == 0, r.stderr[-1500:]
+    doc = _json.loads(out.read_text())
+    assert doc["num_plans"] > 0
+    first = doc["plans"][0]
+    assert first["rank"] == 1 and first["cost_ms"] > 0
+    assert sum(first["device_groups"]) == 8

@@ -384,3 +384,34 @@ def test_het_search_parity_randomized_configs(tmp_path):
              f"{r[1]}, {r[2]}, {r[3]}, {r[4]}") for r in results)
         assert mine == sorted(ref_rows), (
             case, speed, nodes, slots, bw, gbs, len(mine), len(ref_rows))
+
+
+def test_data_balancer_parity_fuzzed():
+    """DataLoadBalancer vs the reference on random hetero DP groups
+    (largest-remainder rounding, per-type throughput bias)."""
+    from metis_amd.planner.balancer import DataLoadBalancer
+    from metis_amd.profiles import LayerProfile, ProfileStore
+
+    rng = random.Random(99)
+    types = ["A100", "V100", "P100", "T4"]
+    cases = []
+    for _ in range(30):
+        dp = rng.choice([2, 3, 4])
+        group = rng.choice([1, 2])
+        seq = [rng.choice(types) for _ in range(dp)]
+        device_types = [t for t in seq for _ in range(group)]
+        bs = rng.randint(dp, 64)
+        totals = {t: round(rng.uniform(5.0, 50.0), 3) for t in set(seq)}
+        cases.append([device_types, dp, 1, bs, totals])
+    ref = probe("data_balancer", {"cases": cases})
+
+    for case, expected in zip(cases, ref):
+        device_types, dp, tp, bs, totals = case
+        store = ProfileStore.__new__(ProfileStore)
+        store._data = {
+            (t, tp, 1): LayerProfile(layer_times_ms=[v],
+                                     layer_memory_mb=[0.0], fb_sync_ms=0.0)
+            for t, v in totals.items()
+        }
+        mine = DataLoadBalancer(store).partition_data(device_types, (dp, tp), bs)
+        assert mine == expected, case
