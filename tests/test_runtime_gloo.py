@@ -497,3 +497,25 @@ def _sp_llama_worker(rank, world, port, out):
 
 def test_sequence_parallel_llama_matches_single():
     _run_workers(_sp_llama_worker, port=29625)
+
+
+def _sp_pp_worker(rank, world, port, out):
+    _env(rank, world, port)
+    from metis_amd.runtime.comm import init_parallel
+    from metis_amd.runtime.runner import PlanRunner
+
+    ctx = init_parallel(dp=1, tp=2, pp=2)
+    torch.manual_seed(5)
+    # sp shrinks the pp boundary tensors to [mbs, s/tp, h]
+    runner = PlanRunner(SPEC, ctx, mbs=1, gbs=2, dtype=torch.float32, sp=True)
+    assert runner.sp
+    loss = runner.train_step()
+    if ctx.is_last_stage:
+        assert loss > 0
+    assert any(p.grad is not None for p in runner.model.parameters())
+    out.put(("ok", rank))
+    dist.destroy_process_group()
+
+
+def test_sp_with_pipeline_boundaries():
+    _run_workers(_sp_pp_worker, world=4, port=29627)
