@@ -71,11 +71,13 @@ def _worker(rank, world, port, out):
     from metis_amd.runtime.comm import init_parallel
     from metis_amd.runtime.runner import PlanRunner
 
+    gbs = int(os.environ.get("IL_GBS", "4"))
+    vpp_i = int(os.environ.get("IL_VPP", "2"))
     ctx = init_parallel(dp=1, tp=1, pp=2)
     results = {}
-    for sched, vpp in (("gpipe", 1), ("interleaved", 2)):
+    for sched, vpp in (("gpipe", 1), ("interleaved", vpp_i)):
         torch.manual_seed(5)
-        runner = PlanRunner(SPEC, ctx, mbs=1, gbs=4, dtype=torch.float32,
+        runner = PlanRunner(SPEC, ctx, mbs=1, gbs=gbs, dtype=torch.float32,
                             schedule=sched, vpp=vpp)
         _det_init(runner)
         loss = runner.train_step()
@@ -86,10 +88,16 @@ def _worker(rank, world, port, out):
     dist.destroy_process_group()
 
 
-def test_interleaved_matches_gpipe():
+def _run_case(port, gbs, vpp):
+    os.environ["IL_GBS"] = str(gbs)
+    os.environ["IL_VPP"] = str(vpp)
+    _launch(port)
+
+
+def _launch(port):
     mp_ctx = mp.get_context("spawn")
     out = mp_ctx.Queue()
-    procs = [mp_ctx.Process(target=_worker, args=(r, 2, 29651, out))
+    procs = [mp_ctx.Process(target=_worker, args=(r, 2, port, out))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -113,3 +121,12 @@ def test_interleaved_matches_gpipe():
     g_g = torch.tensor(got[1][1]["gpipe"][1])
     g_i = torch.tensor(got[0][1]["interleaved"][1])
     assert torch.allclose(g_g, g_i, atol=1e-5), (g_g - g_i).abs().max()
+
+
+def test_interleaved_matches_gpipe_vpp2():
+    _run_case(29652, gbs=4, vpp=2)
+
+
+def test_interleaved_minimal_microbatches():
+    """nm == pp (warmup saturates: all forwards then all backwards)."""
+    _run_case(29653, gbs=2, vpp=2)
