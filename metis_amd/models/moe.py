@@ -60,6 +60,7 @@ class MoEModelSpec:
     top_k: int = 2
     ffn_hidden_size: Optional[int] = None
     aux_loss_coef: float = 0.01
+    expert_activation: str = "gelu"   # "gelu" (GPT FFN) | "swiglu" (Mixtral)
 
     @property
     def ffn(self) -> int:
@@ -75,9 +76,10 @@ class MoEModelSpec:
 
     def num_parameters(self) -> int:
         h, v = self.hidden_size, self.vocab_size
-        per_block = (4 * h * h + 4 * h                       # attention + ln
-                     + self.num_experts * 2 * h * self.ffn   # experts
-                     + h * self.num_experts + 2 * h)         # router + ln
+        fmul = 3 if self.expert_activation == "swiglu" else 2
+        per_block = (4 * h * h + 4 * h                          # attention + ln
+                     + self.num_experts * fmul * h * self.ffn   # experts
+                     + h * self.num_experts + 2 * h)            # router + ln
         return v * h + self.seq_length * h + self.num_layers * per_block + 2 * h
 
 
@@ -85,7 +87,14 @@ MOE_SPECS = {
     # GPT-3 1.3B-shaped attention with 8 experts (~5.4B params, top-2)
     "gpt3-moe-1.3b-8e": MoEModelSpec("gpt3-moe-1.3b-8e", 2048, 24, 16,
                                      51200, 2048, num_experts=8),
+    # Mixtral-8x7B-class shape: swiglu experts, top-2 of 8
+    "mixtral-8x7b-class": MoEModelSpec(
+        "mixtral-8x7b-class", 4096, 32, 32, 32000, 4096, num_experts=8,
+        ffn_hidden_size=14336, expert_activation="swiglu"),
     "moe-tiny": MoEModelSpec("moe-tiny", 64, 2, 4, 512, 32, num_experts=4),
+    "moe-tiny-swiglu": MoEModelSpec("moe-tiny-swiglu", 64, 2, 4, 512, 32,
+                                    num_experts=4,
+                                    expert_activation="swiglu"),
 }
 
 
@@ -98,10 +107,14 @@ class _Experts(nn.Module):
     def __init__(self, spec: MoEModelSpec, ep: int, dtype):
         super().__init__()
         assert spec.num_experts % ep == 0, "num_experts must divide by ep"
+        assert spec.expert_activation in ("gelu", "swiglu")
         self.local_experts = spec.num_experts // ep
+        self.swiglu = spec.expert_activation == "swiglu"
         h, f = spec.hidden_size, spec.ffn
-        self.w1 = nn.Parameter(torch.empty(self.local_experts, f, h, dtype=dtype))
-        self.b1 = nn.Parameter(torch.zeros(self.local_experts, f, dtype=dtype))
+        # swiglu experts fuse gate & up into one [2f, h] weight (Mixtral)
+        f1 = 2 * f if self.swiglu else f
+        self.w1 = nn.Parameter(torch.empty(self.local_experts, f1, h, dtype=dtype))
+        self.b1 = nn.Parameter(torch.zeros(self.local_experts, f1, dtype=dtype))
         self.w2 = nn.Parameter(torch.empty(self.local_experts, h, f, dtype=dtype))
         self.b2 = nn.Parameter(torch.zeros(self.local_experts, h, dtype=dtype))
         for e in range(self.local_experts):
@@ -109,7 +122,12 @@ class _Experts(nn.Module):
             _init_linear(self.w2[e], f)
 
     def expert_forward(self, e: int, x: torch.Tensor) -> torch.Tensor:
-        y = F.gelu(F.linear(x, self.w1[e], self.b1[e]))
+        y = F.linear(x, self.w1[e], self.b1[e])
+        if self.swiglu:
+            gate, up = y.chunk(2, dim=-1)
+            y = F.silu(gate) * up
+        else:
+            y = F.gelu(y)
         return F.linear(y, self.w2[e], self.b2[e])
 
 
@@ -287,8 +305,9 @@ class MoEModel(nn.Module):
         el = 2
         h, v = spec.hidden_size, spec.vocab_size
         embed = (v * h + spec.seq_length * h) * el
+        fmul = 3 if spec.expert_activation == "swiglu" else 2
         per_block = (4 * h * h + 4 * h + 2 * h
-                     + spec.num_experts * (2 * h * spec.ffn + spec.ffn + h)) * el \
+                     + spec.num_experts * (fmul * h * spec.ffn + spec.ffn + h)) * el \
             + h * spec.num_experts * 4  # fp32 router
         head = (v * h + 2 * h) * el
         return [float(embed)] + [float(per_block)] * spec.num_layers + [float(head)]
