@@ -58,6 +58,8 @@ def save_step(runner: PlanRunner, ckpt_dir: str, step: int) -> None:
 
 def main() -> None:
     p = argparse.ArgumentParser()
+    p.add_argument("--config", default=None,
+                   help="YAML file of flag defaults (CLI flags override)")
     p.add_argument("--model", default="gpt2-small", choices=sorted(MODEL_SPECS))
     p.add_argument("--dp", type=int, default=1)
     p.add_argument("--tp", type=int, default=1)
@@ -90,6 +92,15 @@ def main() -> None:
     p.add_argument("--eval-batches", type=int, default=8)
     p.add_argument("--layer-partition", default=None)
     p.add_argument("--log-every", type=int, default=10)
+    cfg_ns, _ = p.parse_known_args()
+    if cfg_ns.config:
+        import yaml
+
+        with open(cfg_ns.config) as fh:
+            doc = yaml.safe_load(fh) or {}
+        unknown = [k for k in doc if not hasattr(cfg_ns, k.replace("-", "_"))]
+        assert not unknown, f"unknown config keys: {unknown}"
+        p.set_defaults(**{k.replace("-", "_"): v for k, v in doc.items()})
     args = p.parse_args()
 
     ctx = init_parallel(dp=args.dp, tp=args.tp, pp=args.pp)

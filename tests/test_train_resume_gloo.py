@@ -78,3 +78,33 @@ def test_latest_complete_step_skips_partial(tmp_path):
     (tmp_path / "step_10" / "COMPLETE").write_text("{}")
     os.makedirs(tmp_path / "step_20")  # no COMPLETE: a rank died mid-write
     assert latest_complete_step(str(tmp_path)) == 10
+
+
+def test_yaml_config_defaults(tmp_path):
+    """--config YAML supplies defaults; explicit CLI flags win."""
+    import subprocess
+    import sys
+
+    import numpy as np
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    np.random.RandomState(0).randint(0, 500, size=32 * 40 + 1).astype(
+        np.uint16).tofile(tmp_path / "t.bin")
+    (tmp_path / "conf.yaml").write_text(
+        f"model: gpt2-small\nsteps: 2\nmbs: 2\ngbs: 2\n"
+        f"data: {tmp_path}/t.bin\nlr: 0.001\nlog-every: 1\n")
+    code = f"""
+import sys; sys.argv = ['train', '--config', r'{tmp_path}/conf.yaml',
+                        '--steps', '1']
+from metis_amd.cli import train
+from metis_amd.models.gpt import GPTModelSpec
+train.MODEL_SPECS = dict(train.MODEL_SPECS)
+train.MODEL_SPECS['gpt2-small'] = GPTModelSpec('tiny', 64, 2, 4, 512, 32)
+train.main()
+"""
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1")
+    r = subprocess.run([sys.executable, "-c", code], cwd=repo, env=env,
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert "step 1:" in r.stdout          # log-every from yaml
+    assert "step 2" not in r.stdout       # --steps 1 overrode yaml's 2
