@@ -219,19 +219,24 @@ class PlanRunner:
         losses: List[torch.Tensor] = []
         aux_terms: List[Optional[torch.Tensor]] = []
 
-        # Pre-post every fill-phase activation recv (and, in reverse order,
-        # every drain-phase grad recv) so p2p transfers land while this
-        # stage computes. GPipe keeps all nm activations live anyway, so
-        # the buffers cost nothing extra; senders emit in this exact order.
+        # On gloo (host-driven) pre-post every fill-phase activation recv
+        # and, in reverse order, every drain-phase grad recv, so p2p lands
+        # while this stage computes (GPipe keeps all nm activations live,
+        # so the buffers are free). On RCCL this would DEADLOCK: p2p ops
+        # to one peer serialize on that peer's internal stream, so drain
+        # recvs posted ahead of the fill sends would block them — the GPU
+        # path keeps recvs in data-flow order.
         nm = self.num_microbatches
+        overlap = (not dist.is_initialized()
+                   or dist.get_backend() == "gloo")
         fwd_bufs, fwd_reqs = [], []
-        if not ctx.is_first_stage:
+        if overlap and not ctx.is_first_stage:
             for _ in range(nm):
                 buf = torch.empty(act_shape, dtype=self.dtype, device=dev)
                 fwd_bufs.append(buf)
                 fwd_reqs.append(dist.irecv(buf, src=prev))
         bwd_bufs, bwd_reqs = {}, {}
-        if not ctx.is_last_stage:
+        if overlap and not ctx.is_last_stage:
             for i in reversed(range(nm)):
                 buf = torch.empty(act_shape, dtype=self.dtype, device=dev)
                 bwd_bufs[i] = buf
@@ -244,8 +249,11 @@ class PlanRunner:
                 x = tokens
                 inputs.append(None)
             else:
-                fwd_reqs[mb].wait()
-                x = fwd_bufs[mb].requires_grad_(True)
+                if overlap:
+                    fwd_reqs[mb].wait()
+                    x = fwd_bufs[mb].requires_grad_(True)
+                else:
+                    x = self._recv_activation(act_shape, prev).requires_grad_(True)
                 inputs.append(x)
             if ctx.is_last_stage:
                 if ctx.is_first_stage:
@@ -269,9 +277,12 @@ class PlanRunner:
                 self.grad_sync.arm()
             if ctx.is_last_stage:
                 (outputs[i] / self.num_microbatches).backward()
-            else:
+            elif overlap:
                 bwd_reqs[i].wait()
                 self._backward_stage(outputs[i], bwd_bufs[i], aux_terms[i])
+            else:
+                gout = self._recv_activation(act_shape, nxt)
+                self._backward_stage(outputs[i], gout, aux_terms[i])
             if not ctx.is_first_stage:
                 dist.send(inputs[i].grad.contiguous(), dst=prev)
 
