@@ -200,6 +200,16 @@ class MoEBlock(nn.Module):
 
         local0 = (dist.get_rank(tp_group) if ep > 1 else 0) * self.experts.local_experts
         out = torch.zeros_like(flat)
+        # every local expert must produce SOME grad every microbatch: a
+        # token-starved expert would otherwise skip its post-accumulate
+        # hook on the final microbatch and desync the DP bucket schedule
+        # across ranks (runtime.grad_sync fails loudly on that). One
+        # element per tensor is enough — autograd materializes the full
+        # (zero) grad.
+        ex = self.experts
+        dummy = (ex.w1.flatten()[0] + ex.b1.flatten()[0]
+                 + ex.w2.flatten()[0] + ex.b2.flatten()[0])
+        out = out + (dummy * 0.0).to(out.dtype)
         for le in range(self.experts.local_experts):
             e = local0 + le
             slot = (top_e == e)                               # [T, k]

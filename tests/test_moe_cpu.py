@@ -93,3 +93,26 @@ def test_swiglu_experts_forward_backward():
     gate, up = F.linear(x, w1, b1).chunk(2, dim=-1)
     ref = F.linear(F.silu(gate) * up, ex.w2[1], ex.b2[1])
     assert torch.allclose(y, ref, atol=1e-6)
+
+
+def test_starved_experts_still_get_grads():
+    """An expert routed zero tokens must still produce a (zero) gradient,
+    or dp grad-sync bucket schedules desync across ranks."""
+    torch.manual_seed(0)
+    m = MoEModel(SPEC, dtype=torch.float32)
+    # rig every router to send all tokens to experts 0 and 1 (top-2)
+    with torch.no_grad():
+        for b in m.blocks:
+            b.router.weight.zero_()
+            b.router.bias.copy_(torch.tensor([10.0, 5.0, -10.0, -10.0]))
+    tokens, labels = _batch()
+    loss = m(tokens, labels=labels)
+    loss.backward()
+    for b in m.blocks:
+        per_expert = b.experts.w1.grad.flatten(1).abs().sum(1)
+        assert per_expert[2] == 0 and per_expert[3] == 0  # truly starved
+        assert b.experts.w1.grad is not None
+        # the hook-relevant property: a grad TENSOR exists for the stacked
+        # expert params even though experts 2/3 saw no tokens
+        assert b.experts.w2.grad is not None
+        assert b.experts.b1.grad is not None
