@@ -70,8 +70,15 @@ def build_app(model, spec, device=None):
 
     @app.post("/generate")
     def gen(req: GenRequest):
+        from fastapi import HTTPException
+
         g = torch.Generator().manual_seed(req.seed) if req.seed else None
+        if not req.tokens or any(len(p) == 0 for p in req.tokens):
+            raise HTTPException(400, "empty prompt")
         lens = {len(p) for p in req.tokens}
+        if max(lens) >= spec.seq_length:
+            raise HTTPException(
+                400, f"prompt length {max(lens)} >= context {spec.seq_length}")
         budget = spec.seq_length - max(lens)
         n = max(0, min(req.max_new_tokens, budget))
         if len(lens) > 1:   # ragged batch: padded-cache batched decode

@@ -77,3 +77,15 @@ def test_ragged_batch_endpoint():
     single = client.post("/generate", json={
         "tokens": [[9, 8]], "max_new_tokens": 4, "temperature": 0.0})
     assert rows[1] == single.json()["tokens"][0]
+
+
+def test_bad_requests_rejected():
+    torch.manual_seed(0)
+    model = GPTModel(SPEC, dtype=torch.float32)
+    model.eval()
+    client = _client(model)
+    assert client.post("/generate", json={"tokens": []}).status_code == 400
+    assert client.post("/generate", json={"tokens": [[]]}).status_code == 400
+    too_long = list(range(SPEC.seq_length + 1))
+    assert client.post(
+        "/generate", json={"tokens": [too_long]}).status_code == 400
