@@ -32,7 +32,6 @@ class GradBucketSync:
         self.opt = optimizer
         self.group = dp_group
         self.dp = dp_size
-        self.enabled = False
         self._handles: List = []
 
         # buckets = contiguous ranges of the optimizer's flat grad buffer,
@@ -58,6 +57,8 @@ class GradBucketSync:
             self._close_bucket(current)
 
         self._pending = [0] * len(self._buckets)
+        self._armed = [False] * len(self.opt.params)
+        self._step_active = False
         for idx, p in enumerate(self.opt.params):
             p.register_post_accumulate_grad_hook(self._make_hook(idx))
 
@@ -71,8 +72,9 @@ class GradBucketSync:
 
     def _make_hook(self, idx: int):
         def hook(param: torch.nn.Parameter) -> None:
-            if not self.enabled:
+            if not self._armed[idx]:
                 return
+            self._armed[idx] = False   # fire once per step per param
             off, n = self.opt._slices[idx]
             flat = self.opt.grad_flat
             g = param.grad
@@ -91,16 +93,33 @@ class GradBucketSync:
 
         return hook
 
-    def arm(self) -> None:
-        """Call before the FINAL microbatch's backward."""
-        self.enabled = True
+    def _begin_step(self) -> None:
+        if self._step_active:
+            return
+        self._step_active = True
         self._handles = []
         for b, (_s, _e, n) in enumerate(self._buckets):
             self._pending[b] = n
 
+    def arm(self) -> None:
+        """Call before the FINAL backward that touches EVERY param (the
+        last microbatch in GPipe/1F1B/no-pipeline)."""
+        self._begin_step()
+        self._armed = [True] * len(self.opt.params)
+
+    def arm_params(self, idxs) -> None:
+        """Arm a subset right before ITS final backward — the interleaved
+        schedule finishes different chunks' gradients at different slots,
+        so chunks arm separately (a bucket spanning two chunks launches
+        once both chunks' members have fired)."""
+        self._begin_step()
+        for i in idxs:
+            self._armed[i] = True
+
     def finish(self) -> None:
         """Wait for outstanding reduces and average; leaves grad_flat ready
         for FusedAdamW.step(pre_gathered=True)."""
+        self._step_active = False
         if any(self._pending):
             # a parameter produced no gradient: ranks would disagree on the
             # bucket schedule — fail loudly rather than deadlock RCCL
@@ -111,5 +130,4 @@ class GradBucketSync:
         for handle, start, end in self._handles:
             handle.wait()
             flat[start:end].div_(self.dp)
-        self.enabled = False
         self._handles = []

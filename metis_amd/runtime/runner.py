@@ -384,6 +384,21 @@ class PlanRunner:
         def b_cm(k):
             return v - 1 - (k // pp) % v, (k // (pp * v)) * pp + k % pp
 
+        # each chunk's grads finish at ITS last backward slot, not the
+        # globally-last one: arm the dp grad-sync per chunk (grad_sync
+        # buckets spanning chunks launch once both sides fired)
+        last_bwd_slot = {}
+        for k in range(total):
+            last_bwd_slot[b_cm(k)[0]] = k
+        if self.grad_sync is not None and not hasattr(self, "_chunk_param_idxs"):
+            idxs = {}
+            pos = 0
+            for c, chunk in enumerate(self.model_chunks):
+                n = sum(1 for _ in chunk.parameters())
+                idxs[c] = list(range(pos, pos + n))
+                pos += n
+            self._chunk_param_idxs = idxs
+
         self.optimizer.zero_grad()
         pending = []
         live = {}
@@ -418,11 +433,11 @@ class PlanRunner:
                     aux = None  # folded into the loss by the chunk itself
             live[(c, mb)] = (inp, out, aux)
 
-        def bwd(k: int, is_last_slot: bool) -> None:
+        def bwd(k: int) -> None:
             c, mb = b_cm(k)
             vs = c * pp + r
-            if self.grad_sync is not None and is_last_slot:
-                self.grad_sync.arm()
+            if self.grad_sync is not None and k == last_bwd_slot[c]:
+                self.grad_sync.arm_params(self._chunk_param_idxs[c])
             inp, out, aux = live.pop((c, mb))
             if vs == last_vs:
                 (out / nm).backward()
@@ -441,11 +456,11 @@ class PlanRunner:
         fk, bk = warmup, 0
         while fk < total:
             fwd(fk)
-            bwd(bk, bk == total - 1)
+            bwd(bk)
             fk += 1
             bk += 1
         while bk < total:
-            bwd(bk, bk == total - 1)
+            bwd(bk)
             bk += 1
         for work, _ in pending:
             work.wait()

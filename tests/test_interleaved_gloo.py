@@ -130,3 +130,38 @@ def test_interleaved_matches_gpipe_vpp2():
 def test_interleaved_minimal_microbatches():
     """nm == pp (warmup saturates: all forwards then all backwards)."""
     _run_case(29653, gbs=2, vpp=2)
+
+
+def _dp_interleaved_worker(rank, world, port, out):
+    _env(rank, world, port)
+    os.environ["METIS_CHECK_SYNC"] = "1"
+    import torch.distributed as dist
+
+    from metis_amd.runtime.comm import init_parallel
+    from metis_amd.runtime.runner import PlanRunner
+
+    ctx = init_parallel(dp=2, tp=1, pp=2)
+    torch.manual_seed(5)
+    runner = PlanRunner(SPEC, ctx, mbs=1, gbs=8, dtype=torch.float32,
+                        schedule="interleaved", vpp=2)
+    # per-chunk arming must complete every DP bucket (a global arm at the
+    # last slot misses earlier chunks' hooks) and keep replicas synced
+    runner.train_step()
+    runner.train_step()
+    out.put(("ok", rank))
+    dist.destroy_process_group()
+
+
+def test_interleaved_with_dp_grad_sync():
+    mp_ctx = mp.get_context("spawn")
+    out = mp_ctx.Queue()
+    procs = [mp_ctx.Process(target=_dp_interleaved_worker,
+                            args=(r, 4, 29654, out)) for r in range(4)]
+    for p in procs:
+        p.start()
+    got = [out.get(timeout=240) for _ in range(4)]
+    for p in procs:
+        p.join(timeout=60)
+    for p in procs:
+        assert p.exitcode == 0, f"worker exit {p.exitcode}"
+    assert len(got) == 4
