@@ -132,6 +132,11 @@ def main() -> None:
                 print(f"resumed from {path} (step {step})")
 
     eval_loader = None
+    if args.eval_every and ctx.pp > 1:
+        # eval runs the pp==1 forward path only; don't skip silently
+        if ctx.rank == 0:
+            print("WARNING: --eval-every is not supported with pp > 1 "
+                  "(eval loss will not be computed)")
     if args.eval_every and args.data and ctx.pp == 1:
         from metis_amd.data import TokenDataset, TokenLoader
 

@@ -28,6 +28,7 @@ from metis_amd.ops.attention import decode_attention, flash_attention
 from metis_amd.ops.cross_entropy import cross_entropy
 from metis_amd.ops.mlp import fused_mlp
 from metis_amd.ops.relayout import heads_merge, qkv_split_transpose
+from metis_amd.partial_grads import mark_partial
 
 import os as _os
 
@@ -467,9 +468,9 @@ class GPTModel(nn.Module):
         self.recompute = False
 
         # SP: params in the seq-sharded regions are replicated but see
-        # only this rank's token slice, so their grads are partial — sum
-        # them across the TP group in backward (same hook pattern as the
-        # MoE router)
+        # only this rank's token slice, so their grads are partial — tag
+        # them for the cross-TP sum (metis_amd.partial_grads: immediate
+        # hook standalone, once-per-step under the runner)
         if self.sp and tp_group is not None:
             hooked = []
             if self.has_embedding:
@@ -483,12 +484,8 @@ class GPTModel(nn.Module):
                 # is the pre-reduce-scatter partial (the head bypasses f)
                 hooked += [self.ln_final.weight, self.ln_final.bias]
 
-            def _sync(p):
-                if p.grad is not None:
-                    dist.all_reduce(p.grad, group=tp_group)
-
             for p in hooked:
-                p.register_post_accumulate_grad_hook(_sync)
+                mark_partial(p, tp_group)
 
     def forward(
         self, x: torch.Tensor, labels: Optional[torch.Tensor] = None,

@@ -26,6 +26,7 @@ from metis_amd.ops.norms import (RMSNorm, apply_rope,
                                  apply_rope_rows, swiglu)
 from metis_amd.ops.relayout import (heads_merge, qkv_rope_split,
                                     qkv_split_transpose)
+from metis_amd.partial_grads import mark_partial
 
 import os as _os
 
@@ -222,7 +223,7 @@ class LlamaModel(nn.Module):
 
         self.recompute = False  # see GPTModel.recompute
 
-        # SP grad-sum hooks for replicated params that see only a seq
+        # SP partial-grad tags for replicated params that see only a seq
         # slice (see GPTModel; llama adds rmsnorm weights + rp biases)
         if self.sp and tp_group is not None:
             hooked = []
@@ -234,12 +235,8 @@ class LlamaModel(nn.Module):
             if self.has_head:
                 hooked.append(self.norm_final.weight)
 
-            def _sync(p):
-                if p.grad is not None:
-                    dist.all_reduce(p.grad, group=tp_group)
-
             for p in hooked:
-                p.register_post_accumulate_grad_hook(_sync)
+                mark_partial(p, tp_group)
 
     def forward(self, x, labels=None, cache=None, pos_offset: int = 0):
         if self.has_embedding:

@@ -16,10 +16,12 @@ Replicated-parameter gradient rules (each rank computes the full router
 but backprops only its local experts' gate terms):
 - expert + router input passes through one f (_CopyToTP), so the
   partial per-rank activation grads are summed to the full grad;
-- router weight/bias grads are summed across the EP group by a
-  post-accumulate-grad hook (tiny tensors: [E, h]);
+- router weight/bias grads are summed across the EP group (tiny
+  tensors: [E, h]) via the metis_amd.partial_grads protocol — an
+  immediate in-backward hook standalone, or exactly once per step on
+  the accumulated grad under the runner's GradBucketSync;
 - the load-balance aux loss is computed identically on every rank and
-  pre-scaled by 1/ep so the hook's sum restores the single logical
+  pre-scaled by 1/ep so the cross-rank sum restores the single logical
   contribution.
 """
 
@@ -46,6 +48,7 @@ from metis_amd.models.gpt import (
 from metis_amd.ops.attention import flash_attention
 from metis_amd.ops.cross_entropy import cross_entropy
 from metis_amd.ops.relayout import heads_merge, qkv_split_transpose
+from metis_amd.partial_grads import mark_partial
 
 
 @dataclass(frozen=True)
@@ -134,7 +137,7 @@ class _Experts(nn.Module):
 class MoEBlock(nn.Module):
     """Pre-LN attention (TP like GPTBlock) + top-k routed expert MLP (EP)."""
 
-    def __init__(self, spec: MoEModelSpec, tp: int, dtype):
+    def __init__(self, spec: MoEModelSpec, tp: int, dtype, tp_group=None):
         super().__init__()
         h = spec.hidden_size
         assert spec.num_heads % tp == 0
@@ -151,31 +154,17 @@ class MoEBlock(nn.Module):
         self.router = nn.Linear(h, spec.num_experts, dtype=torch.float32)
         _init_linear(self.router.weight, h)
         self.experts = _Experts(spec, tp, dtype)
-        self._ep_group = None
-        self._hooked = False
+        # router grads are partial per EP rank; tag them EAGERLY (at
+        # construction, before any GradBucketSync is built) so the EP sum
+        # is ordered before the DP bucket copy — runtime.partial_grads
+        mark_partial(self.router.weight, tp_group)
+        mark_partial(self.router.bias, tp_group)
         self.last_aux_loss: Optional[torch.Tensor] = None
-
-    def _ensure_router_grad_sync(self, group) -> None:
-        """Sum the partial per-rank router grads across the EP group
-        (see module docstring); registered once, fires inside backward."""
-        if self._hooked or group is None or dist.get_world_size(group) <= 1:
-            self._hooked = True
-            return
-        self._ep_group = group
-
-        def _sync(p):
-            if p.grad is not None:
-                dist.all_reduce(p.grad, group=self._ep_group)
-
-        self.router.weight.register_post_accumulate_grad_hook(_sync)
-        self.router.bias.register_post_accumulate_grad_hook(_sync)
-        self._hooked = True
 
     def _moe_mlp(self, y: torch.Tensor, tp_group) -> torch.Tensor:
         spec = self.spec
         b, s, h = y.shape
         ep = dist.get_world_size(tp_group) if tp_group is not None else 1
-        self._ensure_router_grad_sync(tp_group)
 
         flat = _CopyToTP.apply(y, tp_group).reshape(-1, h)    # one f for both uses
         logits = self.router(flat.float())                    # [T, E] replicated
@@ -265,7 +254,7 @@ class MoEModel(nn.Module):
         block_start = max(start - 1, 0)
         block_end = min(end, total - 1) - 1
         self.blocks = nn.ModuleList(
-            MoEBlock(spec, tp, dtype)
+            MoEBlock(spec, tp, dtype, tp_group=tp_group)
             for _ in range(max(block_end - block_start, 0)))
 
         if self.has_head:

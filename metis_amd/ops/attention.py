@@ -4,8 +4,10 @@ Forward: online-softmax flash kernel (attention.hip), saving the row LSE.
 Backward: split dQ / dKV recompute kernels (attention_bwd.hip); GQA dK/dV
 come back per q-head and reduce over the group here (deterministic).
 
-Unsupported shapes (D > 128, D % 16 != 0, S % 64 != 0, non-causal) fall
-back to PyTorch SDPA; on GPU-supported shapes the HIP path always runs.
+Unsupported shapes (D > 128, D % 16 != 0, S % 128 != 0, non-causal) fall
+back to PyTorch SDPA (the fwd/bwd kernels tile the sequence in 128-row
+q blocks — attention.hip QBLK / attention_bwd.hip RBLK); on
+GPU-supported shapes the HIP path always runs.
 """
 
 from __future__ import annotations

@@ -19,6 +19,7 @@ import torch
 import torch.distributed as dist
 
 from metis_amd.ops.adamw import FusedAdamW
+from metis_amd.partial_grads import defer_partial, partial_group
 
 
 class GradBucketSync:
@@ -59,6 +60,11 @@ class GradBucketSync:
         self._pending = [0] * len(self._buckets)
         self._armed = [False] * len(self.opt.params)
         self._step_active = False
+        # replicated-with-partial-grads params (MoE router over EP, SP
+        # norm/bias params over TP): their cross-rank sum moves INTO our
+        # armed hook — before the bucket copy, once per step, on the
+        # fully-accumulated grad (runtime.partial_grads protocol)
+        defer_partial(self.opt.params)
         for idx, p in enumerate(self.opt.params):
             p.register_post_accumulate_grad_hook(self._make_hook(idx))
 
@@ -75,6 +81,11 @@ class GradBucketSync:
             if not self._armed[idx]:
                 return
             self._armed[idx] = False   # fire once per step per param
+            pg = partial_group(param)
+            if pg is not None and param.grad is not None:
+                # partial-grad sum (EP/TP) before the flat-buffer copy the
+                # DP all-reduce, clipping and FusedAdamW consume
+                dist.all_reduce(param.grad, group=pg)
             off, n = self.opt._slices[idx]
             flat = self.opt.grad_flat
             g = param.grad

@@ -28,6 +28,7 @@ from metis_amd.planner.volume import uniform_layer_split
 from metis_amd.runtime.clip import global_grad_norm, shard_flags
 from metis_amd.runtime.comm import ParallelContext
 from metis_amd.runtime.grad_sync import GradBucketSync
+from metis_amd.partial_grads import defer_partial, sync_partial_grads
 from metis_amd.runtime.trace import tracer_from_env
 
 
@@ -141,6 +142,12 @@ class PlanRunner:
         self.grad_sync = None
         if ctx.dp > 1 and ctx.dp_group is not None:
             self.grad_sync = GradBucketSync(self.optimizer, ctx.dp_group, ctx.dp)
+        else:
+            # the runner accumulates over microbatches: replicated
+            # partial-grad params (MoE router / SP norms) must sum across
+            # their group once per step, not per backward — defer the
+            # model-level hooks and do it in _sync_and_step
+            defer_partial(self.optimizer.params)
 
     # --- data -------------------------------------------------------------
     def synthetic_batch(self) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -485,6 +492,7 @@ class PlanRunner:
                 # overlapped with backward; wait and average
                 self.grad_sync.finish()
             else:
+                sync_partial_grads(self.optimizer.params)
                 self.optimizer.gather_grads()
         scale = 1.0
         if self.clip_grad is not None:

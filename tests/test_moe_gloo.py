@@ -165,3 +165,78 @@ def _moe_interleaved_worker(rank, world, port, out):
 
 def test_moe_interleaved_schedule():
     _run_workers(_moe_interleaved_worker, port=29626)
+
+
+def _ep_accum_worker(rank, world, port, out):
+    """Gradient accumulation over 2 microbatches with ep=2:
+    the deferred once-per-step EP sum must reproduce the full model's
+    accumulated router grad — the old per-backward hook re-summed earlier
+    microbatches (3x instead of 2x on identical data)."""
+    _env(rank, world, port)
+    from metis_amd.runtime.comm import init_parallel
+    from metis_amd.partial_grads import defer_partial, sync_partial_grads
+
+    ctx = init_parallel(dp=1, tp=2, pp=1)
+    torch.manual_seed(7)
+    full = MoEModel(SPEC, tp=1, dtype=torch.float32)
+    shard = MoEModel(SPEC, tp=2, dtype=torch.float32, tp_group=ctx.tp_group)
+    _shard_from_full(full, shard, rank, 2)
+
+    g = torch.Generator().manual_seed(13)
+    mbs = [(torch.randint(0, SPEC.vocab_size, (2, 32), generator=g))
+           for _ in range(2)]
+
+    # accumulation driver contract: defer hooks, sum once after the last
+    # backward (what PlanRunner/GradBucketSync do internally)
+    defer_partial(shard.parameters())
+    for tokens in mbs:
+        labels = torch.roll(tokens, -1, 1)
+        (full(tokens, labels=labels) / 2).backward()
+        (shard(tokens, labels=labels) / 2).backward()
+    sync_partial_grads(shard.parameters())
+
+    fr = full.blocks[0].router.weight.grad
+    sr = shard.blocks[0].router.weight.grad
+    assert torch.allclose(fr, sr, atol=1e-4), (fr - sr).abs().max()
+    fb = full.blocks[-1].router.bias.grad
+    sb = shard.blocks[-1].router.bias.grad
+    assert torch.allclose(fb, sb, atol=1e-4)
+    out.put(("ok", rank))
+    dist.destroy_process_group()
+
+
+def test_ep_grad_accumulation_once_per_step():
+    _run_workers(_ep_accum_worker, port=29631)
+
+
+def _dp_ep_worker(rank, world, port, out):
+    """dp=2 x ep=2 with accumulation: the EP sum must run
+    BEFORE the DP bucket copy (it now happens inside the bucket hook), so
+    router weights stay identical across TP ranks after optimizer steps."""
+    _env(rank, world, port)
+    from metis_amd.runtime.comm import init_parallel
+    from metis_amd.runtime.runner import PlanRunner
+
+    ctx = init_parallel(dp=2, tp=2, pp=1)
+    torch.manual_seed(5)
+    runner = PlanRunner(SPEC, ctx, mbs=1, gbs=4, dtype=torch.float32)
+    assert runner.num_microbatches == 2     # accumulation active
+    assert runner.grad_sync is not None     # DP bucket path active
+    for _ in range(2):
+        runner.train_step()
+
+    # replicated router params must be bit-identical across the WORLD
+    # (both across DP replicas and across TP/EP ranks)
+    for blk in runner.model.blocks:
+        for p in (blk.router.weight, blk.router.bias):
+            flat = p.detach().reshape(-1)
+            gathered = [torch.empty_like(flat) for _ in range(world)]
+            dist.all_gather(gathered, flat)
+            for gth in gathered[1:]:
+                assert torch.equal(gathered[0], gth)
+    out.put(("ok", rank))
+    dist.destroy_process_group()
+
+
+def test_dp_ep_router_stays_synced():
+    _run_workers(_dp_ep_worker, world=4, port=29632)
