@@ -54,6 +54,17 @@ class PlannerArgs:
     # "alpha_beta" — t = alpha + bytes / BW with measured latency term
     comm_model: str = "parity"
     alpha_us: float = 20.0          # per-collective latency when comm_model="alpha_beta"
+    # "parity"   — fb_sync residual charged once per MICROBATCH
+    #              (reference cost_estimator.py:120); overprices gradient
+    #              accumulation (+20% measured at gpt2-small mbs=2).
+    # "marginal" — per-microbatch time = measured accumulation marginal
+    #              (profile extension keys fwd_bwd_{1,2}mb_ms), iteration
+    #              residual charged ONCE; falls back to parity for
+    #              profiles without the keys.
+    microbatch_model: str = "parity"
+    # Linearly interpolate layer times/memory/fb_sync between profiled
+    # batch sizes instead of skipping unprofiled-mbs plans (KeyError).
+    interpolate_bs: bool = False
     activation_dtype_bytes: int = 1  # 1 => element-count parity (quirk Q8); 2 for bf16 bytes
     # The reference's LayerComputeBalancer can emit partitions that do NOT
     # cover every layer on skewed many-stage inputs (slice rounding drops a

@@ -68,6 +68,24 @@ class CostEstimatorBase:
     def _oom(self, stage_memory_mb: Sequence[float]) -> bool:
         return self.cluster.device_memory_mb(0) < max(stage_memory_mb)
 
+    def _profile(self, dtype: str, tp: int, bs: int):
+        """Profile lookup honoring the interpolate_bs extension."""
+        if self.args.interpolate_bs:
+            return self.profiles.get_interp(dtype, tp, bs)
+        return self.profiles.get(dtype, tp, bs)
+
+    def _use_marginal(self, prof) -> bool:
+        return (self.args.microbatch_model == "marginal"
+                and prof.marginal_mb_ms is not None)
+
+    def _fb_sync_of(self, prof) -> float:
+        """fb_sync of a (possibly interpolated) profile; a 0.0 value
+        raises KeyError like a missing one — reference parity
+        (cost_estimator.py:68-69, quirk Q15)."""
+        if not prof.fb_sync_ms:
+            raise KeyError("fb_sync missing (or 0.0) in profile data")
+        return prof.fb_sync_ms
+
 
 class HomoCostEstimator(CostEstimatorBase):
     """Uniform-plan estimator (cost_estimator.py:83-138)."""
@@ -94,14 +112,21 @@ class HomoCostEstimator(CostEstimatorBase):
             start = sum(stage_layers[:stage_id])
             end = sum(stage_layers[: stage_id + 1])
 
-            prof = self.profiles.get(device_type, tp, bs)
-            lens.append(prof.time_slice(start, end))
+            prof = self._profile(device_type, tp, bs)
+            if self._use_marginal(prof):
+                # per-microbatch stage time = measured accumulation
+                # marginal; iteration residual charged ONCE (summed over
+                # stage slices = whole-model residual)
+                lens.append(prof.marginal_slice(start, end))
+                fb_sync_cost += prof.residual_slice(start, end)
+            else:
+                lens.append(prof.time_slice(start, end))
+                if stage_id == len(stage_layers) - 1:
+                    fb_sync_cost = self._fb_sync_of(prof) * num_mbs
             stage_params.append(sum(param_sizes[start:end]))
             stage_memory.append(prof.memory_slice(start, end))
 
-            if stage_id == len(stage_layers) - 1:
-                fb_sync_cost = self.profiles.max_fb_sync([device_type], tp, bs) * num_mbs
-            else:
+            if stage_id != len(stage_layers) - 1:
                 act = self.volume.activation_size(end, bs, tp)
                 pp_bw = self.topology.slowest_pp_bandwidth((pp, tp, dp), stage_id)
                 pp_cost += self._pp_cost(act, pp_bw)
@@ -136,7 +161,9 @@ class HeteroCostEstimator(CostEstimatorBase):
     ) -> float:
         dp_deg, tp_deg = strategy
         if len(set(device_types)) == 1:
-            prof = self.profiles.get(device_types[0], tp_deg, gbs // dp_deg // batches)
+            prof = self._profile(device_types[0], tp_deg, gbs // dp_deg // batches)
+            if self._use_marginal(prof):
+                return prof.marginal_slice(start_layer, end_layer)
             return prof.time_slice(start_layer, end_layer)
 
         balancer = DataLoadBalancer(self.profiles)
@@ -150,7 +177,11 @@ class HeteroCostEstimator(CostEstimatorBase):
             for bs_slice in pow2_slices(h_bs):
                 if bs_slice > self.args.max_profiled_batch_size:
                     raise KeyError(f"batch_size({bs_slice}) not profiled")
-                t += self.profiles.get(dtype, tp_deg, bs_slice).time_slice(start_layer, end_layer)
+                prof = self._profile(dtype, tp_deg, bs_slice)
+                if self._use_marginal(prof):
+                    t += prof.marginal_slice(start_layer, end_layer)
+                else:
+                    t += prof.time_slice(start_layer, end_layer)
             costs.append(t)
         return max(costs)
 
@@ -181,9 +212,21 @@ class HeteroCostEstimator(CostEstimatorBase):
 
             dp_deg, tp_deg = strategy
             mbs = plan.gbs // dp_deg // plan.batches
-            if stage_id == plan.num_stage - 1:
-                fb_sync_cost = self.profiles.max_fb_sync(device_types, tp_deg, mbs) * plan.batches
+            if self.args.microbatch_model == "marginal":
+                profs = [self._profile(d, tp_deg, mbs)
+                         for d in set(device_types)]
             else:
+                profs = []
+            if profs and all(self._use_marginal(p) for p in profs):
+                # iteration residual charged once per stage slice (the
+                # slowest device type's), not per microbatch
+                fb_sync_cost += max(
+                    p.residual_slice(start_l, end_l) for p in profs)
+            elif stage_id == plan.num_stage - 1:
+                fb_sync_cost = max(
+                    self._fb_sync_of(self._profile(d, tp_deg, mbs))
+                    for d in set(device_types)) * plan.batches
+            if stage_id != plan.num_stage - 1:
                 act = self.volume.activation_size(end_l, mbs, tp_deg)
                 pp_cost += self._pp_cost(act, topology.slowest_pp_bandwidth(stage_id))
 

@@ -251,6 +251,36 @@ def profile_model(
     for t in timers:
         t.remove()
 
+    # hook-free accumulation probe: fwd+bwd of a 1- vs 2-microbatch
+    # gradient-accumulation iteration at this bs. The difference is the
+    # steady-state cost of one accumulated microbatch (extension keys
+    # fwd_bwd_1mb_ms / fwd_bwd_2mb_ms -> LayerProfile.marginal_mb_ms);
+    # the hooked fwd_bwd above keeps the reference's per-iteration
+    # semantics (README.md:174-186)
+    def fwd_bwd_k(kmb: int) -> float:
+        opt.zero_grad()
+        tokens, labels = batch()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(kmb):
+            x = embed(tokens)
+            for blk in model.blocks:
+                if recompute:
+                    x = torch.utils.checkpoint.checkpoint(
+                        blk, x, model.tp_group, use_reentrant=False)
+                else:
+                    x = blk(x, model.tp_group)
+            (head(x, labels) / kmb).backward()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) * 1000
+
+    for k in (1, 2):
+        fwd_bwd_k(k)   # warm each variant
+    probe_iters = max(iters // 2, 3)
+    t1mb = sum(fwd_bwd_k(1) for _ in range(probe_iters)) / probe_iters
+    t2mb = sum(fwd_bwd_k(2) for _ in range(probe_iters)) / probe_iters
+    opt.zero_grad()
+
     rank = dist.get_rank() if dist.is_initialized() else 0
     if rank != 0:
         return None
@@ -269,6 +299,8 @@ def profile_model(
         layer_compute_total_ms=layer_ms,
         total_memory_mb=sum(layer_mem_mb),
         layer_memory_total_mb=layer_mem_mb,
+        fwd_bwd_1mb_ms=t1mb,
+        fwd_bwd_2mb_ms=t2mb,
     )
     return path
 

@@ -36,17 +36,43 @@ from typing import Dict, List, Optional, Tuple
 
 @dataclass
 class LayerProfile:
-    """Measurements for one (device_type, tp, bs) point."""
+    """Measurements for one (device_type, tp, bs) point.
+
+    ``marginal_mb_ms``/``residual_ms`` come from the MI355X profiler's
+    extension keys ``fwd_bwd_1mb_ms``/``fwd_bwd_2mb_ms`` (hook-free
+    fwd+bwd of a 1- vs 2-microbatch gradient-accumulation iteration):
+    ``marginal`` is the steady-state cost of one accumulated microbatch
+    (t2 - t1) and ``residual`` the once-per-iteration remainder
+    (t1 - marginal). The reference schema has no such keys — its
+    fb_sync residual is charged once per MICROBATCH
+    (cost_estimator.py:120), which overprices accumulation (measured
+    +20% at gpt2-small mbs=2); the marginal model fixes that while the
+    parity path keeps reference behavior."""
 
     layer_times_ms: List[float]      # fwd+bwd per layer
     layer_memory_mb: List[float]     # per layer
     fb_sync_ms: float                # residual sync cost per microbatch
+    marginal_mb_ms: Optional[float] = None   # per extra accum microbatch
+    residual_ms: Optional[float] = None      # once-per-iteration remainder
 
     def time_slice(self, start: int, end: int) -> float:
         return sum(self.layer_times_ms[start:end])
 
     def memory_slice(self, start: int, end: int) -> float:
         return sum(self.layer_memory_mb[start:end])
+
+    def marginal_slice(self, start: int, end: int) -> float:
+        """Per-accumulated-microbatch time of layers [start, end): the
+        measured whole-model marginal scaled by this slice's share of
+        the per-layer compute."""
+        total = sum(self.layer_times_ms)
+        share = self.time_slice(start, end) / total if total else 0.0
+        return self.marginal_mb_ms * share
+
+    def residual_slice(self, start: int, end: int) -> float:
+        total = sum(self.layer_times_ms)
+        share = self.time_slice(start, end) / total if total else 0.0
+        return self.residual_ms * share
 
 
 @dataclass
@@ -109,10 +135,17 @@ class ProfileStore:
     def add_raw(self, dtype: str, tp: int, bs: int, raw: dict) -> None:
         et = raw["execution_time"]
         layer_times = [float(t) for t in et["layer_compute_total_ms"]]
+        marginal = residual = None
+        if "fwd_bwd_1mb_ms" in et and "fwd_bwd_2mb_ms" in et:
+            t1, t2 = float(et["fwd_bwd_1mb_ms"]), float(et["fwd_bwd_2mb_ms"])
+            marginal = max(t2 - t1, 0.0)
+            residual = max(t1 - marginal, 0.0)
         prof = LayerProfile(
             layer_times_ms=layer_times,
             layer_memory_mb=[float(m) for m in raw["execution_memory"]["layer_memory_total_mb"]],
             fb_sync_ms=float(et["forward_backward_time_ms"]) - sum(layer_times),
+            marginal_mb_ms=marginal,
+            residual_ms=residual,
         )
         self._data[(dtype, tp, bs)] = prof
         if dtype not in self.device_type_names:
@@ -139,6 +172,42 @@ class ProfileStore:
         if key not in self._data:
             raise KeyError(f"profile tp{tp}_bs{bs} for {dtype} not found")
         return self._data[key]
+
+    def get_interp(self, dtype: str, tp: int, bs: int) -> LayerProfile:
+        """Like get(), but for an unprofiled bs strictly inside the
+        profiled range, linearly interpolate every time/memory quantity
+        between the bracketing profiled batch sizes (MI355X extension;
+        the reference skips such plans outright, cost_het_cluster.py:46).
+        Still raises KeyError outside the profiled range — extrapolation
+        is not trustworthy."""
+        if (dtype, tp, bs) in self._data:
+            return self._data[(dtype, tp, bs)]
+        sizes = sorted(b for (d, t, b) in self._data if d == dtype and t == tp)
+        lo = max((b for b in sizes if b < bs), default=None)
+        hi = min((b for b in sizes if b > bs), default=None)
+        if lo is None or hi is None:
+            raise KeyError(
+                f"profile tp{tp}_bs{bs} for {dtype} not found and not "
+                f"bracketed by profiled sizes {sizes}")
+        p_lo, p_hi = self._data[(dtype, tp, lo)], self._data[(dtype, tp, hi)]
+        w = (bs - lo) / (hi - lo)
+
+        def mix(a: float, b: float) -> float:
+            return a + (b - a) * w
+
+        both = (p_lo.marginal_mb_ms is not None
+                and p_hi.marginal_mb_ms is not None)
+        return LayerProfile(
+            layer_times_ms=[mix(a, b) for a, b in
+                            zip(p_lo.layer_times_ms, p_hi.layer_times_ms)],
+            layer_memory_mb=[mix(a, b) for a, b in
+                             zip(p_lo.layer_memory_mb, p_hi.layer_memory_mb)],
+            fb_sync_ms=mix(p_lo.fb_sync_ms, p_hi.fb_sync_ms),
+            marginal_mb_ms=mix(p_lo.marginal_mb_ms, p_hi.marginal_mb_ms)
+            if both else None,
+            residual_ms=mix(p_lo.residual_ms, p_hi.residual_ms)
+            if both else None,
+        )
 
     def fb_sync(self, dtype: str, tp: int, bs: int) -> float:
         """fb_sync of one point; a 0.0 value raises KeyError like a missing
@@ -170,6 +239,8 @@ class ProfileStore:
         layer_compute_total_ms: List[float],
         total_memory_mb: float,
         layer_memory_total_mb: List[float],
+        fwd_bwd_1mb_ms: Optional[float] = None,
+        fwd_bwd_2mb_ms: Optional[float] = None,
     ) -> None:
         doc = {
             "model": {
@@ -194,5 +265,11 @@ class ProfileStore:
                 "layer_memory_total_mb": layer_memory_total_mb,
             },
         }
+        if fwd_bwd_1mb_ms is not None and fwd_bwd_2mb_ms is not None:
+            # MI355X extension keys (hook-free 1- vs 2-microbatch
+            # accumulation iteration) — see LayerProfile docstring;
+            # unknown to the reference loader, which ignores extra keys
+            doc["execution_time"]["fwd_bwd_1mb_ms"] = fwd_bwd_1mb_ms
+            doc["execution_time"]["fwd_bwd_2mb_ms"] = fwd_bwd_2mb_ms
         with open(path, "w") as fh:
             json.dump(doc, fh, indent=2)

@@ -101,3 +101,118 @@ def test_fb_sync_scales_with_batch(tiny_setup):
     _, store, *_ = tiny_setup
     assert store.fb_sync("MI355X", 1, 1) == pytest.approx(1.0)
     assert store.fb_sync("MI355X", 1, 2) == pytest.approx(2.0)
+
+
+@pytest.fixture()
+def marginal_setup(tmp_path):
+    """Like tiny_setup but the profiles carry the MI355X accumulation
+    extension keys (fwd_bwd_{1,2}mb_ms): marginal = 7*bs, residual = 2*bs.
+    Profiled bs: 1 and 4 (so bs=2 exercises interpolation)."""
+    prof_dir = tmp_path / "prof"
+    prof_dir.mkdir()
+    for bs in (1, 4):
+        scale = bs
+        ProfileStore.write_profile_json(
+            str(prof_dir / f"DeviceType.MI355X_tp1_bs{bs}.json"),
+            model_name="tiny",
+            parameters_per_layer_bytes=[100.0, 50.0, 50.0, 100.0],
+            total_time_ms=10.0 * scale,
+            forward_backward_time_ms=9.0 * scale,
+            batch_generator_time_ms=0.5,
+            layernorm_grads_all_reduce_time_ms=0.0,
+            embedding_grads_all_reduce_time_ms=0.0,
+            optimizer_time_ms=4.0,
+            layer_compute_total_ms=[1.0 * scale, 2.0 * scale, 2.0 * scale,
+                                    3.0 * scale],
+            total_memory_mb=80.0,
+            layer_memory_total_mb=[10.0, 20.0, 20.0, 30.0],
+            fwd_bwd_1mb_ms=9.0 * scale,
+            fwd_bwd_2mb_ms=16.0 * scale,
+        )
+    (tmp_path / "hostfile").write_text("a slots=2\nb slots=2\n")
+    (tmp_path / "hostfile1").write_text("a slots=1\n")
+    (tmp_path / "clusterfile.json").write_text(json.dumps({
+        "a": {"instance_type": "MI355X", "inter_bandwidth": 10,
+              "intra_bandwidth": 100, "memory": 288},
+        "b": {"instance_type": "MI355X", "inter_bandwidth": 10,
+              "intra_bandwidth": 100, "memory": 288},
+    }))
+    cluster = ClusterSpec(str(tmp_path / "hostfile"),
+                          str(tmp_path / "clusterfile.json"))
+    cluster1 = ClusterSpec(str(tmp_path / "hostfile1"),
+                           str(tmp_path / "clusterfile.json"))
+    store = ProfileStore.load_dir(str(prof_dir), optimizer_scale=1.0)
+    cfg = ModelConfig("tiny", 4, 8, 16, 32)
+    vol = GPTVolume(cfg, store.model.parameters_per_layer_bytes)
+    return (cluster, cluster1), store, cfg, vol
+
+
+def test_marginal_microbatch_model_by_hand(marginal_setup):
+    (cluster, cluster1), store, cfg, vol = marginal_setup
+    # marginal = 16-9 = 7, residual = 9-7 = 2 (at bs=1)
+    prof = store.get("MI355X", 1, 1)
+    assert prof.marginal_mb_ms == pytest.approx(7.0)
+    assert prof.residual_ms == pytest.approx(2.0)
+
+    est1 = HomoCostEstimator(store, cfg, vol, cluster1,
+                             PlannerArgs(gbs=4, max_profiled_tp_degree=1,
+                                         max_profiled_batch_size=4,
+                                         microbatch_model="marginal"))
+    est = HomoCostEstimator(store, cfg, vol, cluster,
+                            PlannerArgs(gbs=4, max_profiled_tp_degree=1,
+                                        max_profiled_batch_size=4,
+                                        microbatch_model="marginal"))
+    # dp=1, pp=1, mbs=1, gbs=4 -> 4 microbatches
+    cost, _, _ = est1.get_cost(UniformPlan(1, 1, 1, 1, 4), "MI355X")
+    # exec = (4-1)*7 + 7 = 28; residual ONCE = 2; opt = 4; bg = 0.5*4
+    assert cost == pytest.approx(28 + 2 + 4 + 0.5 * 4)
+
+    # dp=2, pp=2, mbs=1, gbs=4 -> 2 microbatches; lc shares [3/8, 5/8]
+    cost2, _, _ = est.get_cost(UniformPlan(2, 2, 1, 1, 4), "MI355X")
+    lens = [7 * 3 / 8, 7 * 5 / 8]
+    exec2 = (2 - 1) * max(lens) + sum(lens)
+    pp_term = 128 / (100 * 1024 * 1024)
+    dp_term = 2 * 1 / (2 * 100 * 1024 * 1024) * 150
+    assert cost2 == pytest.approx(exec2 + 2 + 2 + pp_term + dp_term + 0.5 * 2)
+
+    # parity mode ignores the extension keys: fb_sync charged per mb
+    est_p = HomoCostEstimator(store, cfg, vol, cluster1,
+                              PlannerArgs(gbs=4, max_profiled_tp_degree=1,
+                                          max_profiled_batch_size=4))
+    cost_p, _, _ = est_p.get_cost(UniformPlan(1, 1, 1, 1, 4), "MI355X")
+    assert cost_p == pytest.approx(4 * 8 + 1 * 4 + 4 + 0.5 * 4)
+
+
+def test_bs_interpolation(marginal_setup):
+    (cluster, cluster1), store, cfg, vol = marginal_setup
+    # bs=2 is not profiled: get raises, get_interp blends bs1 and bs4
+    with pytest.raises(KeyError):
+        store.get("MI355X", 1, 2)
+    prof = store.get_interp("MI355X", 1, 2)
+    w = (2 - 1) / (4 - 1)
+    assert prof.layer_times_ms[0] == pytest.approx(1 + (4 - 1) * w)
+    assert prof.fb_sync_ms == pytest.approx(1 + (4 - 1) * w)
+    assert prof.marginal_mb_ms == pytest.approx(7 + (28 - 7) * w)
+    # outside the range still raises
+    with pytest.raises(KeyError):
+        store.get_interp("MI355X", 1, 8)
+    # bs=3 IS bracketed -> interpolates
+    prof3 = store.get_interp("MI355X", 1, 3)
+    assert prof3.layer_times_ms[-1] == pytest.approx(3 + (12 - 3) * 2 / 3)
+
+    # estimator path: mbs=2 plan costs without KeyError when enabled
+    est = HomoCostEstimator(store, cfg, vol, cluster1,
+                            PlannerArgs(gbs=8, max_profiled_tp_degree=1,
+                                        max_profiled_batch_size=4,
+                                        microbatch_model="marginal",
+                                        interpolate_bs=True))
+    cost, _, _ = est.get_cost(UniformPlan(1, 1, 1, 2, 8), "MI355X")
+    marg2 = 7 + (28 - 7) * w
+    res2 = 2 + (8 - 2) * w
+    assert cost == pytest.approx((4 - 1) * marg2 + marg2 + res2 + 4 + 0.5 * 4)
+    # disabled -> plans at unprofiled mbs are skipped (reference behavior)
+    est_off = HomoCostEstimator(store, cfg, vol, cluster1,
+                                PlannerArgs(gbs=8, max_profiled_tp_degree=1,
+                                            max_profiled_batch_size=4))
+    with pytest.raises(KeyError):
+        est_off.get_cost(UniformPlan(1, 1, 1, 2, 8), "MI355X")
