@@ -73,6 +73,7 @@ def main() -> None:
 
     profile_dir = args.profile_dir or f"profiles/mi355x/{args.model}"
     est_ms = None
+    mc = None
     if os.path.isdir(profile_dir):
         from metis_amd.config import ModelConfig
         from metis_amd.cli.plan_search import best_plan, estimate_plan
@@ -89,13 +90,33 @@ def main() -> None:
                               comm_bench_path="profiles/comm_bench.json")
             if found:
                 dp, tp, pp, mbs, est_ms = found
-        if est_ms is None:
-            # report the cost-model estimate for the plan we are running
-            est_ms = estimate_plan(profile_dir, mc, n_gpus, gbs,
-                                   dp=dp, tp=tp, pp=pp, mbs=mbs,
-                                   comm_bench_path="profiles/comm_bench.json")
 
     ctx = init_parallel(dp=dp, tp=tp, pp=pp)
+
+    # At N > 1, calibrate the xGMI comm constants IN-PROCESS before the
+    # timed region (a short rccl-tests-style all-reduce sweep over the
+    # world group) and feed the measured bus bandwidth + latency into the
+    # cost-model estimate — the planner's comm terms are then measured,
+    # not clusterfile guesses. Reported in config for the record.
+    comm_busbw = comm_alpha = None
+    if n_gpus > 1 and torch.cuda.is_available() and dist.is_initialized():
+        from metis_amd.profiler.comm_bench import bench_allreduce
+
+        rows = bench_allreduce([1 << 16, 1 << 22, 1 << 26, 1 << 28],
+                               iters=8, warmup=3)
+        comm_busbw = round(max(r["busbw_GBps"] for r in rows), 1)
+        comm_alpha = round(min(r["time_us"] for r in rows), 1)
+
+    if mc is not None and est_ms is None:
+        from metis_amd.cli.plan_search import estimate_plan
+
+        # report the cost-model estimate for the plan we are running
+        est_ms = estimate_plan(profile_dir, mc, n_gpus, gbs,
+                               dp=dp, tp=tp, pp=pp, mbs=mbs,
+                               comm_bench_path="profiles/comm_bench.json",
+                               intra_bandwidth=comm_busbw,
+                               alpha_us=comm_alpha,
+                               schedule=args.schedule, vpp=args.vpp)
     runner = PlanRunner(spec, ctx, mbs=mbs, gbs=gbs,
                         schedule=args.schedule, recompute=args.recompute,
                         zero1=args.zero1, sp=args.sp, vpp=args.vpp)
@@ -135,6 +156,8 @@ def main() -> None:
                 "cost_model_error_pct": (
                     abs(est_ms - ms) / ms * 100.0 if est_ms else None
                 ),
+                "comm_busbw_GBps": comm_busbw,
+                "comm_alpha_us": comm_alpha,
             },
         }
         print(json.dumps(result))

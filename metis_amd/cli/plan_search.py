@@ -50,9 +50,13 @@ def best_plan(
     max_tp: int = 8,
     max_bs: int = 16,
     comm_bench_path: Optional[str] = None,
+    intra_bandwidth: Optional[float] = None,
+    alpha_us: Optional[float] = None,
 ) -> Optional[Tuple[int, int, int, int, float]]:
     """Returns (dp, tp, pp, mbs, est_ms) of the cheapest feasible plan, or
-    None when no profiled plan fits."""
+    None when no profiled plan fits. Explicit ``intra_bandwidth``/
+    ``alpha_us`` (e.g. measured in-process by bench.py) override the
+    comm_bench.json suggestion."""
     intra = 130.0
     alpha = 20.0
     if comm_bench_path and os.path.exists(comm_bench_path):
@@ -60,12 +64,19 @@ def best_plan(
             sugg = json.load(fh).get("clusterfile_suggestion", {})
         intra = sugg.get("intra_bandwidth", intra)
         alpha = sugg.get("alpha_us", alpha)
+    if intra_bandwidth is not None:
+        intra = intra_bandwidth
+    if alpha_us is not None:
+        alpha = alpha_us
 
     cluster = single_node_cluster(n_gpus, device_type, intra_bandwidth=intra)
     store = ProfileStore.load_dir(profile_dir, optimizer_scale=1.0)
+    # MI355X flow: measured-accumulation microbatch model + bs
+    # interpolation (falls back to parity on profiles without the keys)
     args = PlannerArgs(gbs=gbs, max_profiled_tp_degree=max_tp,
                        max_profiled_batch_size=max_bs,
-                       comm_model="alpha_beta", alpha_us=alpha)
+                       comm_model="alpha_beta", alpha_us=alpha,
+                       microbatch_model="marginal", interpolate_bs=True)
     results = search_homo_cluster(cluster, store, model_config, args,
                                   device_type=device_type)
     feasible = [(p, c) for p, c, oom in results if not oom]
@@ -87,6 +98,10 @@ def estimate_plan(
     mbs: int,
     device_type: str = "MI355X",
     comm_bench_path: Optional[str] = None,
+    intra_bandwidth: Optional[float] = None,
+    alpha_us: Optional[float] = None,
+    schedule: str = "gpipe",
+    vpp: int = 1,
 ) -> Optional[float]:
     """Cost-model estimate (ms) for one specific plan; None if unprofiled."""
     from metis_amd.planner.cost import HomoCostEstimator
@@ -99,6 +114,10 @@ def estimate_plan(
             sugg = json.load(fh).get("clusterfile_suggestion", {})
         intra = sugg.get("intra_bandwidth", intra)
         alpha = sugg.get("alpha_us", alpha)
+    if intra_bandwidth is not None:
+        intra = intra_bandwidth
+    if alpha_us is not None:
+        alpha = alpha_us
     try:
         cluster = single_node_cluster(n_gpus, device_type, intra_bandwidth=intra)
         store = ProfileStore.load_dir(profile_dir, optimizer_scale=1.0)
@@ -107,7 +126,9 @@ def estimate_plan(
             store, model_config, volume, cluster,
             PlannerArgs(gbs=gbs, max_profiled_tp_degree=max(tp, 8),
                         max_profiled_batch_size=max(mbs, 16),
-                        comm_model="alpha_beta", alpha_us=alpha),
+                        comm_model="alpha_beta", alpha_us=alpha,
+                        microbatch_model="marginal", interpolate_bs=True,
+                        schedule=schedule, vpp=vpp),
         )
         cost, _, _ = est.get_cost(UniformPlan(dp, pp, tp, mbs, gbs), device_type)
         return cost

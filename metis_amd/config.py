@@ -65,6 +65,14 @@ class PlannerArgs:
     # Linearly interpolate layer times/memory/fb_sync between profiled
     # batch sizes instead of skipping unprofiled-mbs plans (KeyError).
     interpolate_bs: bool = False
+    # Pipeline schedule the homo estimator prices (the runtime executes
+    # all three). "gpipe" keeps the reference bubble (B-1)*max + sum
+    # (cost_estimator.py:129); "1f1b" has the same bubble but holds at
+    # most pp microbatches of activations in flight instead of B;
+    # "interleaved" with vpp virtual chunks per stage shrinks the bubble
+    # to (pp-1)/vpp microbatch slots at vpp x the p2p volume.
+    schedule: str = "gpipe"
+    vpp: int = 1
     activation_dtype_bytes: int = 1  # 1 => element-count parity (quirk Q8); 2 for bf16 bytes
     # The reference's LayerComputeBalancer can emit partitions that do NOT
     # cover every layer on skewed many-stage inputs (slice rounding drops a

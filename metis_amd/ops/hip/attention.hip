@@ -27,7 +27,18 @@ typedef short short4v __attribute__((ext_vector_type(4)));
 
 constexpr int THREADS = 256;
 constexpr int QBLK = 128;     // q rows per block (2 x 16 per wave)
-constexpr int VPAD = 8;       // Vt row padding (bf16 elements)
+constexpr int PROW = 20;      // P^T LDS row length (16 q + 4 pad shorts):
+                              // 40-B row stride keeps the packed b64 P
+                              // stores conflict-free (10*i mod 32 distinct
+                              // over a 16-lane group) and tr16 reads
+                              // 8-B aligned
+// defer-max threshold (T13, log2 domain): skip the O/l rescale while the
+// running max grows by < THR — P is then bounded by 2^THR instead of 1,
+// which the fp32 l/O accumulators absorb (P itself is bf16: ~3x looser
+// abs error on O, still ~1e-2-class for unit-variance inputs)
+constexpr float RESCALE_THR = 8.0f;
+constexpr float LOG2E = 1.4426950408889634f;
+constexpr float LN2 = 0.6931471805599453f;
 
 // KV tile width: 128 for D <= 80 (the K/V^T/P LDS images still fit two
 // blocks per CU), 64 above. Wider tiles amortize the two barriers per
@@ -54,7 +65,6 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
     constexpr int dchunks = (D + 31) / 32;   // 32-wide K-dim chunks of D
     constexpr int djtiles = D / 16;          // 16-wide output column tiles
     constexpr int KSLOT = D / 8 + 1;         // K LDS slots per row (padded)
-    constexpr int VROW = KVBLK + VPAD;       // Vt LDS row length
 
     const int qtile = blockIdx.x % (S / QBLK);
     const int head = (blockIdx.x / (S / QBLK)) % H;
@@ -73,10 +83,14 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
     // numerically — never store element-wise through the struct type).
     // V stays in its NATURAL [kv][D] image: PV B-fragments read it with
     // ds_read_b64_tr_b16 (hardware transpose), so no scalar transpose
-    // staging exists anywhere in this kernel.
+    // staging exists anywhere in this kernel. P is staged TRANSPOSED
+    // ([kv][16 q], one image per wave, reused across the wave's two row
+    // blocks): the C-fragment's 4 q values per lane are one packed
+    // ds_write_b64, and the PV A-fragment comes back via the same tr16
+    // hardware-transpose read the V path uses — no scalar LDS traffic.
     short* Ks = reinterpret_cast<short*>(smem);                 // [KVBLK][KSLOT*8]
     short* Vs = Ks + KVBLK * KSLOT * 8;                         // [KVBLK][KSLOT*8]
-    short* Pw = Vs + KVBLK * KSLOT * 8 + wave * 2 * 16 * VROW;  // [2][16][VROW]
+    short* Pw = Vs + KVBLK * KSLOT * 8 + wave * KVBLK * PROW;   // [KVBLK][PROW]
 
     // ---- preload Q fragments for both row blocks ------------------------
     bf16x8 q_frag[2][dchunks];
@@ -161,15 +175,17 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
         if (kv0 + KVBLK < kv_end) issue_loads(kv0 + KVBLK);
         __syncthreads();
 
-        // ---- per row block: S = scale * Q K^T, online softmax, P -> LDS -
-        bool rb_active[2];
-        float p[2][JSUB][4];   // [rb][j][r]
+        // ---- per row block: S = scale * Q K^T, online softmax, P^T ->
+        //      LDS (packed), O += P V — one fused pass per block so the
+        //      single per-wave P^T image can be reused between the two
+        const int p4 = lane & 15;
         #pragma unroll
         for (int rb = 0; rb < 2; ++rb) {
             const int q0 = qb + rbid[rb] * 16;
-            rb_active[rb] = kv0 <= q0 + 15;
-            if (!rb_active[rb]) continue;
+            if (kv0 > q0 + 15) continue;
 
+            float p[JSUB][4];   // [j][r], this lane: q rows k8*4+r,
+                                // kv col j*16+col16 (C fragment layout)
             __builtin_amdgcn_s_setprio(1);   // T5: favor the MFMA cluster
             #pragma unroll
             for (int j = 0; j < JSUB; ++j) {    // 16-col subtiles
@@ -193,69 +209,92 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
                 for (int r = 0; r < 4; ++r) {
                     const int qrow = q0 + k8 * 4 + r;
                     const int kvcol = kv0 + j * 16 + col16;
-                    float sv = s_acc[r] * scale;
-                    p[rb][j][r] = (kvcol > qrow) ? -1e30f : sv;
+                    // log2 domain: v_exp_f32 natively computes 2^x, so
+                    // folding log2(e) into the scale deletes one multiply
+                    // per element of every later exp
+                    float sv = s_acc[r] * (scale * LOG2E);
+                    p[j][r] = (kvcol > qrow) ? -1e30f : sv;
                 }
             }
             __builtin_amdgcn_s_setprio(0);
 
+            // T13 defer-max: only rescale O/l when some row's max grew
+            // past THR (in log2 units); P stays bounded by 2^THR
+            float tile_max[4];
+            bool need = false;
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                float tile_max = -1e30f;
+                float tm = p[0][r];
                 #pragma unroll
-                for (int j = 0; j < JSUB; ++j)
-                    tile_max = fmaxf(tile_max, p[rb][j][r]);
+                for (int j = 1; j < JSUB; ++j) tm = fmaxf(tm, p[j][r]);
                 #pragma unroll
                 for (int off = 8; off > 0; off >>= 1)
-                    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, off, 16));
-
-                const float m_new = fmaxf(m_run[rb][r], tile_max);
-                const float alpha = __expf(m_run[rb][r] - m_new);
+                    tm = fmaxf(tm, __shfl_xor(tm, off, 16));
+                tile_max[r] = tm;
+                need = need || (tm > m_run[rb][r] + RESCALE_THR);
+            }
+            if (__any(need)) {
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const float m_new = fmaxf(m_run[rb][r], tile_max[r]);
+                    const float alpha = exp2f(m_run[rb][r] - m_new);
+                    m_run[rb][r] = m_new;
+                    l_run[rb][r] *= alpha;
+                    #pragma unroll
+                    for (int jd = 0; jd < djtiles; ++jd)
+                        o_acc[rb][jd][r] *= alpha;
+                }
+            }
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
                 float row_sum = 0.f;
                 #pragma unroll
                 for (int j = 0; j < JSUB; ++j) {
-                    p[rb][j][r] = __expf(p[rb][j][r] - m_new);
-                    row_sum += p[rb][j][r];
+                    p[j][r] = exp2f(p[j][r] - m_run[rb][r]);
+                    row_sum += p[j][r];
                 }
                 #pragma unroll
                 for (int off = 8; off > 0; off >>= 1)
                     row_sum += __shfl_xor(row_sum, off, 16);
-
-                l_run[rb][r] = l_run[rb][r] * alpha + row_sum;
-                m_run[rb][r] = m_new;
-                #pragma unroll
-                for (int jd = 0; jd < djtiles; ++jd)
-                    o_acc[rb][jd][r] *= alpha;
+                l_run[rb][r] += row_sum;
             }
 
-            short* Prb = Pw + rb * 16 * VROW;
+            // P^T image: row kv = j*16+col16, cols q = k8*4..+3 — the
+            // lane's 4 C-fragment values are one packed 8-B store
             #pragma unroll
-            for (int j = 0; j < JSUB; ++j)
+            for (int j = 0; j < JSUB; ++j) {
+                short4v pk;
                 #pragma unroll
                 for (int r = 0; r < 4; ++r)
-                    Prb[(k8 * 4 + r) * VROW + j * 16 + col16] =
-                        float_to_bf16_bits(p[rb][j][r]);
-        }
+                    pk[r] = float_to_bf16_bits(p[j][r]);
+                *reinterpret_cast<short4v*>(
+                    Pw + (j * 16 + col16) * PROW + k8 * 4) = pk;
+            }
 
-        // P is per-wave private: DS ops of one wave complete in order, so
-        // a compiler-level fence (no barrier) suffices to keep the vector
-        // re-read below the scalar writes above.
-        asm volatile("" ::: "memory");
+            // P is per-wave private: DS ops of one wave complete in
+            // order, so a compiler-level fence (no barrier) keeps the
+            // tr16 re-reads below the writes above.
+            asm volatile("" ::: "memory");
 
-        // ---- O += P @ V -------------------------------------------------
-        #pragma unroll
-        for (int rb = 0; rb < 2; ++rb) {
-            if (!rb_active[rb]) continue;
-            const short* Prb = Pw + rb * 16 * VROW;
-            // V B-fragments via hardware transpose read: each 16-lane
-            // group cooperatively loads one 4(kv)x16(d) block — lane
-            // l&15 receives its d-column's 4 kv values. Per fragment,
-            // two tr16 reads cover the 8 kv rows of this lane group.
-            const int p4 = lane & 15;     // piece index within the group
+            // ---- O += P @ V ---------------------------------------------
+            // Both A (P^T image) and B (V natural image) fragments come
+            // via tr16 hardware-transpose reads: each 16-lane group
+            // loads a 4(kv)x16 block, lane l&15 receiving its column's
+            // 4 kv values.
             #pragma unroll
             for (int ks = 0; ks < KVBLK / 32; ++ks) {   // 32-wide kv chunks
-                bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(
-                    Prb + col16 * VROW + ks * 32 + k8 * 8);
+                bf16x8 p_frag;
+                #pragma unroll
+                for (int r = 0; r < 2; ++r) {
+                    const int kvrow = ks * 32 + k8 * 8 + 4 * r + (p4 >> 2);
+                    short4v t = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                        (__attribute__((address_space(3))) short4v*)(
+                            Pw + kvrow * PROW + (p4 & 3) * 4));
+                    p_frag[r * 4 + 0] = t[0];
+                    p_frag[r * 4 + 1] = t[1];
+                    p_frag[r * 4 + 2] = t[2];
+                    p_frag[r * 4 + 3] = t[3];
+                }
                 #pragma unroll
                 for (int jd = 0; jd < djtiles; ++jd) {
                     bf16x8 v_frag;
@@ -291,8 +330,10 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
                 O[q_base + (long)qrow * D + jd * 16 + col16] =
                     __float2bfloat16(o_acc[rb][jd][r] * inv_l);
             if (col16 == 0)
+                // m_run lives in the log2 domain (see QK^T scale fold);
+                // LSE stays natural-log for the backward pass
                 LSE[((long)batch * H + head) * S + qrow] =
-                    m_run[rb][r] + __logf(l_run[rb][r]);
+                    m_run[rb][r] * LN2 + __logf(l_run[rb][r]);
         }
     }
 }
@@ -319,7 +360,7 @@ std::vector<torch::Tensor> attn_fwd(
         do {                                                                  \
             const int kvb = kvblk_for<DD>();                                  \
             const int lds = (2 * kvb * (DD / 8 + 1) * 8                       \
-                             + 4 * 2 * 16 * (kvb + VPAD)) * 2;                \
+                             + 4 * kvb * PROW) * 2;                           \
             hipLaunchKernelGGL(attn_fwd_kernel<DD>, dim3(grid),               \
                 dim3(THREADS), lds, stream,                                   \
                 reinterpret_cast<const bf16*>(qc.data_ptr()),                 \

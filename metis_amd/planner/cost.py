@@ -131,8 +131,34 @@ class HomoCostEstimator(CostEstimatorBase):
                 pp_bw = self.topology.slowest_pp_bandwidth((pp, tp, dp), stage_id)
                 pp_cost += self._pp_cost(act, pp_bw)
 
+        # --- schedule pricing (MI355X extension; default keeps parity) ---
+        # gpipe : (B-1)*max + sum (reference cost_estimator.py:129), one
+        #         microbatch of activations charged (reference memory
+        #         semantics).
+        # 1f1b  : same bubble, but activation memory scales with the
+        #         in-flight microbatch count min(B, pp - stage_id).
+        # interleaved (vpp=v): bubble shrinks to (pp-1)/v chunk slots —
+        #         execution = (B-1)*max + (sum + (v-1)*max)/v, which
+        #         reduces to B*t + (pp-1)*t/v on uniform stages — at v x
+        #         the p2p boundary crossings; in-flight count as 1f1b.
+        sched, v = self.args.schedule, max(self.args.vpp, 1)
+        if sched != "gpipe":
+            state_mb = [p * 9.0 / (1024 * 1024) for p in
+                        self.volume.parameter_sizes(tp)]
+            for sid in range(len(stage_layers)):
+                start = sum(stage_layers[:sid])
+                end = sum(stage_layers[: sid + 1])
+                state = sum(state_mb[start:end])
+                act = max(stage_memory[sid] - state, 0.0)
+                inflight = min(num_mbs, pp - sid) if pp > 1 else 1
+                stage_memory[sid] = state + act * max(inflight, 1)
         oom = self._oom(stage_memory)
-        execution = (num_mbs - 1) * max(lens) + sum(lens)
+        if sched == "interleaved" and pp > 1:
+            execution = ((num_mbs - 1) * max(lens)
+                         + (sum(lens) + (v - 1) * max(lens)) / v)
+            pp_cost *= v
+        else:
+            execution = (num_mbs - 1) * max(lens) + sum(lens)
         optimizer = self.profiles.model.optimizer_time_ms / pp / tp
         dp_bw = self.topology.slowest_dp_bandwidth((pp, tp, dp))
         dp_cost = self._dp_cost(stage_params, dp_bw, dp)
