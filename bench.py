@@ -78,12 +78,19 @@ def main() -> None:
         from metis_amd.config import ModelConfig
         from metis_amd.cli.plan_search import best_plan, estimate_plan
 
+        moe = {}
+        if hasattr(spec, "num_experts"):
+            moe = dict(num_experts=spec.num_experts,
+                       expert_weight_mul=(
+                           3 if spec.expert_activation == "swiglu" else 2),
+                       ffn_hidden_size=spec.ffn)
         mc = ModelConfig(
             model_name=spec.name,
             num_layers=spec.profile_num_layers,
             hidden_size=spec.hidden_size,
             sequence_length=spec.seq_length,
             vocab_size=spec.vocab_size,
+            **moe,
         )
         if args.plan_search:
             found = best_plan(profile_dir, mc, n_gpus, gbs,

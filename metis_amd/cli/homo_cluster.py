@@ -18,7 +18,7 @@ from metis_amd.planner.cost import HomoCostEstimator
 from metis_amd.planner.plans import UniformPlan
 from metis_amd.planner.uniform import uniform_plans
 from metis_amd.planner.validate import CostValidator, plan_key
-from metis_amd.planner.volume import GPTVolume
+from metis_amd.planner.volume import make_volume
 from metis_amd.profiles import ProfileStore
 
 log = logging.getLogger(__name__)
@@ -33,7 +33,7 @@ def search_homo_cluster(
 ) -> List[Tuple[UniformPlan, float, bool]]:
     """Sweep uniform plans at gbs == planner_args.gbs; returns
     (plan, cost_ms, oom) tuples (unprofiled points skipped)."""
-    volume = GPTVolume(
+    volume = make_volume(
         model_config,
         profiles.model.parameters_per_layer_bytes,
         planner_args.activation_dtype_bytes,

@@ -106,7 +106,7 @@ def estimate_plan(
     """Cost-model estimate (ms) for one specific plan; None if unprofiled."""
     from metis_amd.planner.cost import HomoCostEstimator
     from metis_amd.planner.plans import UniformPlan
-    from metis_amd.planner.volume import GPTVolume
+    from metis_amd.planner.volume import make_volume
 
     intra, alpha = 130.0, 20.0
     if comm_bench_path and os.path.exists(comm_bench_path):
@@ -121,7 +121,7 @@ def estimate_plan(
     try:
         cluster = single_node_cluster(n_gpus, device_type, intra_bandwidth=intra)
         store = ProfileStore.load_dir(profile_dir, optimizer_scale=1.0)
-        volume = GPTVolume(model_config, store.model.parameters_per_layer_bytes)
+        volume = make_volume(model_config, store.model.parameters_per_layer_bytes)
         est = HomoCostEstimator(
             store, model_config, volume, cluster,
             PlannerArgs(gbs=gbs, max_profiled_tp_degree=max(tp, 8),

@@ -29,6 +29,12 @@ class ModelConfig:
     ffn_hidden_size: Optional[int] = None   # defaults to 4*hidden for GPT
     num_attention_heads: Optional[int] = None
     num_kv_heads: Optional[int] = None      # GQA (Llama); None => MHA
+    # MoE (EP axis, absent in the reference): >0 selects MoEVolume in
+    # the planner — experts shard over the EP(=TP) group while router +
+    # norms stay replicated, which changes the per-rank parameter bytes
+    # the DP ring-all-reduce and memory terms price.
+    num_experts: int = 0
+    expert_weight_mul: int = 2              # 2 = gelu FFN, 3 = swiglu
 
     def __post_init__(self) -> None:
         if self.ffn_hidden_size is None:

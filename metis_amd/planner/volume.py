@@ -64,6 +64,63 @@ class GPTVolume:
         return total
 
 
+class MoEVolume(GPTVolume):
+    """MoE (EP axis) volume model — an MI355X extension with no
+    reference counterpart (the reference models dense GPT only,
+    model/activation_parameter.py).
+
+    EP degree == TP degree in the runtime (models/moe.py): experts
+    shard across the group while the router (fp32) and norms stay
+    replicated on every rank, so per-rank parameter bytes are
+    ``replicated + (total - replicated) / tp`` rather than the dense
+    ``total / tp``. Activations crossing stage boundaries are the dense
+    ``bs*seq*hidden`` (the top-k expert combine happens inside the
+    block), so activation_size is inherited.
+    """
+
+    def __init__(
+        self,
+        model_config: ModelConfig,
+        parameters_per_layer_bytes: Sequence[float],
+        activation_dtype_bytes: int = 1,
+    ) -> None:
+        super().__init__(model_config, parameters_per_layer_bytes,
+                         activation_dtype_bytes)
+        h = model_config.hidden_size
+        # replicated per block: fp32 router [E, h] + bias, 2 LayerNorms
+        # (w+b), row-parallel proj bias — bytes
+        self.block_replicated = float(
+            (h * model_config.num_experts + model_config.num_experts) * 4
+            + (4 * h + h) * 2)
+        self.block_replicated = min(self.block_replicated,
+                                    self.transformer_params)
+
+    def parameter_sizes(self, tp_deg: int) -> List[float]:
+        n = self.config.num_layers
+        block = (self.block_replicated
+                 + (self.transformer_params - self.block_replicated) / tp_deg)
+        # embedding replicated across TP in the runtime; head sharded
+        return ([self.input_params]
+                + [block] * (n - 2)
+                + [self.output_params / tp_deg])
+
+    def stage_parameter_size(self, tp_deg: int, start_layer: int,
+                             end_layer: int) -> float:
+        sizes = self.parameter_sizes(tp_deg)
+        return float(sum(sizes[start_layer:end_layer]))
+
+
+def make_volume(
+    model_config: ModelConfig,
+    parameters_per_layer_bytes: Sequence[float],
+    activation_dtype_bytes: int = 1,
+):
+    """GPTVolume for dense configs, MoEVolume when num_experts > 0."""
+    cls = MoEVolume if model_config.num_experts > 0 else GPTVolume
+    return cls(model_config, parameters_per_layer_bytes,
+               activation_dtype_bytes)
+
+
 def uniform_layer_split(total_layers: int, num_stages: int) -> List[int]:
     """Even split of (total-2) transformer layers over stages, remainder to
     stages 1..r, +1 (embed/head) on first and last stage.

@@ -262,3 +262,24 @@ def test_schedule_aware_pricing(marginal_setup):
                                                 (4, 3, 2, 1)))]
     for got, want in zip(mem_f, expect):
         assert got == pytest.approx(want)
+
+
+def test_moe_volume_parameter_split():
+    """MoEVolume: experts shard over EP(=tp), router+norms replicated,
+    embedding replicated, head sharded."""
+    from metis_amd.planner.volume import MoEVolume, make_volume
+    cfg = ModelConfig("moe", 4, 64, 32, 512, num_experts=4,
+                      ffn_hidden_size=256)
+    params = [1000.0, 4000.0, 4000.0, 2000.0]
+    vol = make_volume(cfg, params)
+    assert isinstance(vol, MoEVolume)
+    repl = (64 * 4 + 4) * 4 + (4 * 64 + 64) * 2
+    sizes = vol.parameter_sizes(2)
+    assert sizes[0] == 1000.0                       # embedding replicated
+    assert sizes[-1] == 1000.0                      # head / tp
+    assert sizes[1] == pytest.approx(repl + (4000.0 - repl) / 2)
+    # dense config still yields GPTVolume with /tp everywhere
+    dense = make_volume(ModelConfig("g", 4, 64, 32, 512), params)
+    assert dense.parameter_sizes(2)[0] == 500.0
+    # stage sum consistent
+    assert vol.stage_parameter_size(2, 0, 4) == pytest.approx(sum(sizes))
