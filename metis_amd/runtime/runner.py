@@ -226,28 +226,26 @@ class PlanRunner:
         losses: List[torch.Tensor] = []
         aux_terms: List[Optional[torch.Tensor]] = []
 
-        # On gloo (host-driven) pre-post every fill-phase activation recv
-        # and, in reverse order, every drain-phase grad recv, so p2p lands
-        # while this stage computes (GPipe keeps all nm activations live,
-        # so the buffers are free). On RCCL this would DEADLOCK: p2p ops
-        # to one peer serialize on that peer's internal stream, so drain
-        # recvs posted ahead of the fill sends would block them — the GPU
-        # path keeps recvs in data-flow order.
+        # Pre-posted, order-safe p2p overlap (both gloo AND RCCL):
+        # p2p ops between one pair of ranks complete in issue order, so
+        # overlap is safe exactly when both ends issue their ops for a
+        # pair in the same sequence. Fill recvs (from prev) are posted
+        # up-front — prev sends them in the same 0..nm-1 order. Drain
+        # recvs (from nxt) are posted only AFTER the last fill send to
+        # nxt, in reverse order — matching nxt's grad-send order. (The
+        # round-1 version pre-posted drain recvs at loop top, which is
+        # only safe on gloo's host-driven transport; on RCCL they would
+        # sit ahead of the fill sends in the pair's issue order.)
+        # GPipe keeps all nm activations live anyway, so the buffers are
+        # free; sends are isend with the tensor kept alive until waited.
         nm = self.num_microbatches
-        overlap = (not dist.is_initialized()
-                   or dist.get_backend() == "gloo")
+        pending_sends: List[Tuple] = []
         fwd_bufs, fwd_reqs = [], []
-        if overlap and not ctx.is_first_stage:
+        if not ctx.is_first_stage:
             for _ in range(nm):
                 buf = torch.empty(act_shape, dtype=self.dtype, device=dev)
                 fwd_bufs.append(buf)
                 fwd_reqs.append(dist.irecv(buf, src=prev))
-        bwd_bufs, bwd_reqs = {}, {}
-        if overlap and not ctx.is_last_stage:
-            for i in reversed(range(nm)):
-                buf = torch.empty(act_shape, dtype=self.dtype, device=dev)
-                bwd_bufs[i] = buf
-                bwd_reqs[i] = dist.irecv(buf, src=nxt)
 
         # forward fill
         for mb in range(nm):
@@ -256,11 +254,8 @@ class PlanRunner:
                 x = tokens
                 inputs.append(None)
             else:
-                if overlap:
-                    fwd_reqs[mb].wait()
-                    x = fwd_bufs[mb].requires_grad_(True)
-                else:
-                    x = self._recv_activation(act_shape, prev).requires_grad_(True)
+                fwd_reqs[mb].wait()
+                x = fwd_bufs[mb].requires_grad_(True)
                 inputs.append(x)
             if ctx.is_last_stage:
                 if ctx.is_first_stage:
@@ -276,7 +271,17 @@ class PlanRunner:
                 out = self.model(x)
                 outputs.append(out)
                 aux_terms.append(self._stage_aux())
-                dist.send(out.detach().contiguous(), dst=nxt)
+                t = out.detach().contiguous()
+                pending_sends.append((dist.isend(t, dst=nxt), t))
+
+        # drain-phase grad recvs: issue order to nxt is now
+        # send(0..nm-1) then recv(nm-1..0) on both ends
+        bwd_bufs, bwd_reqs = {}, {}
+        if not ctx.is_last_stage:
+            for i in reversed(range(nm)):
+                buf = torch.empty(act_shape, dtype=self.dtype, device=dev)
+                bwd_bufs[i] = buf
+                bwd_reqs[i] = dist.irecv(buf, src=nxt)
 
         # backward drain (reverse order)
         for i in reversed(range(self.num_microbatches)):
@@ -284,15 +289,15 @@ class PlanRunner:
                 self.grad_sync.arm()
             if ctx.is_last_stage:
                 (outputs[i] / self.num_microbatches).backward()
-            elif overlap:
+            else:
                 bwd_reqs[i].wait()
                 self._backward_stage(outputs[i], bwd_bufs[i], aux_terms[i])
-            else:
-                gout = self._recv_activation(act_shape, nxt)
-                self._backward_stage(outputs[i], gout, aux_terms[i])
             if not ctx.is_first_stage:
-                dist.send(inputs[i].grad.contiguous(), dst=prev)
+                t = inputs[i].grad.contiguous()
+                pending_sends.append((dist.isend(t, dst=prev), t))
 
+        for work, _t in pending_sends:
+            work.wait()
         self._sync_and_step()
         if losses:
             return float(torch.stack([l.detach() for l in losses]).mean())
