@@ -81,13 +81,15 @@ def decode_attention(q: torch.Tensor, k: torch.Tensor,
                      v: torch.Tensor) -> torch.Tensor:
     """Single-query attention over cached K/V (serving decode step):
     q [B, H, 1, D] vs k/v [B, Hkv, S, D], GQA mapped in-kernel. The
-    gfx950 kernel (decode.hip) runs under METIS_DECODE_KERNEL=1 until
-    GPU-validated; otherwise SDPA over the (expanded) cache."""
+    gfx950 kernel (decode.hip) is the DEFAULT on supported shapes —
+    GPU-validated round 2 (numerics test + measured llama3-1b decode
+    317 -> 295 ms for 8x64 tokens, +7.4% tok/s, gpurun_out/r2ab);
+    METIS_DECODE_KERNEL=0 forces the SDPA fallback."""
     import os
 
     if (q.is_cuda and q.dtype == torch.bfloat16 and q.size(2) == 1
             and q.size(-1) in (64, 80, 96, 128)
-            and os.environ.get("METIS_DECODE_KERNEL") == "1"):
+            and os.environ.get("METIS_DECODE_KERNEL", "1") == "1"):
         ext = _ops.require_extension()
         return ext.attn_decode(q, k, v, 1.0 / math.sqrt(q.size(-1)))
     if k.size(1) != q.size(1):

@@ -250,9 +250,9 @@ def test_fp8_matmul_matches_bf16_within_quant_tolerance():
 
 
 @pytest.mark.gpu
-@pytest.mark.skipif(os.environ.get("METIS_EXPERIMENTAL") != "1",
-                    reason="decode attention kernel pending GPU validation")
 def test_attn_decode_matches_sdpa():
+    # default path since round 2 (decode_attention flips it on unless
+    # METIS_DECODE_KERNEL=0): always numerics-tested
     import math
 
     import metis_amd._hip_ops as ext
