@@ -274,11 +274,15 @@ def profile_model(
         torch.cuda.synchronize()
         return (time.perf_counter() - t0) * 1000
 
-    for k in (1, 2):
+    # (t4 - t2)/2 isolates the steady-state accumulation marginal: both
+    # arms are in the accumulate-grad regime and launch-pipelined, which
+    # a 2-vs-1 probe is not (measured: it overestimates the marginal by
+    # ~25% on launch-bound gpt2-small bs1 and puts the residual at 0)
+    for k in (2, 4):
         fwd_bwd_k(k)   # warm each variant
     probe_iters = max(iters // 2, 3)
-    t1mb = sum(fwd_bwd_k(1) for _ in range(probe_iters)) / probe_iters
     t2mb = sum(fwd_bwd_k(2) for _ in range(probe_iters)) / probe_iters
+    t4mb = sum(fwd_bwd_k(4) for _ in range(probe_iters)) / probe_iters
     opt.zero_grad()
 
     rank = dist.get_rank() if dist.is_initialized() else 0
@@ -299,8 +303,8 @@ def profile_model(
         layer_compute_total_ms=layer_ms,
         total_memory_mb=sum(layer_mem_mb),
         layer_memory_total_mb=layer_mem_mb,
-        fwd_bwd_1mb_ms=t1mb,
         fwd_bwd_2mb_ms=t2mb,
+        fwd_bwd_4mb_ms=t4mb,
     )
     return path
 

@@ -39,11 +39,13 @@ class LayerProfile:
     """Measurements for one (device_type, tp, bs) point.
 
     ``marginal_mb_ms``/``residual_ms`` come from the MI355X profiler's
-    extension keys ``fwd_bwd_1mb_ms``/``fwd_bwd_2mb_ms`` (hook-free
-    fwd+bwd of a 1- vs 2-microbatch gradient-accumulation iteration):
-    ``marginal`` is the steady-state cost of one accumulated microbatch
-    (t2 - t1) and ``residual`` the once-per-iteration remainder
-    (t1 - marginal). The reference schema has no such keys — its
+    extension keys: hook-free fwd+bwd of k-microbatch gradient-
+    accumulation iterations. Preferred probe ``fwd_bwd_{2,4}mb_ms``:
+    ``marginal`` = (t4 - t2)/2 — both arms in the steady accumulate-grad
+    regime — and ``residual`` = t2 - 2*marginal, the once-per-iteration
+    remainder. (Legacy ``fwd_bwd_{1,2}mb_ms`` files use t2 - t1, which
+    overestimates launch-bound small-model marginals by ~25%.) The
+    reference schema has no such keys — its
     fb_sync residual is charged once per MICROBATCH
     (cost_estimator.py:120), which overprices accumulation (measured
     +20% at gpt2-small mbs=2); the marginal model fixes that while the
@@ -136,7 +138,12 @@ class ProfileStore:
         et = raw["execution_time"]
         layer_times = [float(t) for t in et["layer_compute_total_ms"]]
         marginal = residual = None
-        if "fwd_bwd_1mb_ms" in et and "fwd_bwd_2mb_ms" in et:
+        if "fwd_bwd_2mb_ms" in et and "fwd_bwd_4mb_ms" in et:
+            # preferred probe: both arms in the steady accumulation regime
+            t2, t4 = float(et["fwd_bwd_2mb_ms"]), float(et["fwd_bwd_4mb_ms"])
+            marginal = max((t4 - t2) / 2, 0.0)
+            residual = max(t2 - 2 * marginal, 0.0)
+        elif "fwd_bwd_1mb_ms" in et and "fwd_bwd_2mb_ms" in et:
             t1, t2 = float(et["fwd_bwd_1mb_ms"]), float(et["fwd_bwd_2mb_ms"])
             marginal = max(t2 - t1, 0.0)
             residual = max(t1 - marginal, 0.0)
@@ -241,6 +248,7 @@ class ProfileStore:
         layer_memory_total_mb: List[float],
         fwd_bwd_1mb_ms: Optional[float] = None,
         fwd_bwd_2mb_ms: Optional[float] = None,
+        fwd_bwd_4mb_ms: Optional[float] = None,
     ) -> None:
         doc = {
             "model": {
@@ -265,11 +273,14 @@ class ProfileStore:
                 "layer_memory_total_mb": layer_memory_total_mb,
             },
         }
-        if fwd_bwd_1mb_ms is not None and fwd_bwd_2mb_ms is not None:
-            # MI355X extension keys (hook-free 1- vs 2-microbatch
-            # accumulation iteration) — see LayerProfile docstring;
-            # unknown to the reference loader, which ignores extra keys
+        # MI355X extension keys (hook-free k-microbatch accumulation
+        # iterations) — see LayerProfile docstring; unknown to the
+        # reference loader, which ignores extra keys
+        if fwd_bwd_1mb_ms is not None:
             doc["execution_time"]["fwd_bwd_1mb_ms"] = fwd_bwd_1mb_ms
+        if fwd_bwd_2mb_ms is not None:
             doc["execution_time"]["fwd_bwd_2mb_ms"] = fwd_bwd_2mb_ms
+        if fwd_bwd_4mb_ms is not None:
+            doc["execution_time"]["fwd_bwd_4mb_ms"] = fwd_bwd_4mb_ms
         with open(path, "w") as fh:
             json.dump(doc, fh, indent=2)
