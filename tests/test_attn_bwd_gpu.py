@@ -79,3 +79,27 @@ def test_model_trains_with_flash_kernels():
         opt.step()
         losses.append(float(loss.detach()))
     assert losses[-1] < losses[0], losses
+
+
+def test_flash_attention_forward_rescale_branch_forced():
+    """Deferred-max (T13) branch test, cdna guide rule 26: a spiked K row
+    at a late tile forces the running max to jump past RESCALE_THR, so
+    the rescale path (not just the defer path) is exercised; compare the
+    full tensor against an fp32 reference."""
+    torch.manual_seed(2)
+    b, h, s, d = 1, 4, 1024, 128
+    q = torch.randn(b, h, s, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, h, s, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, h, s, d, device="cuda", dtype=torch.bfloat16)
+    # spike in tile 4 (kv 520) and a weaker one in tile 7 (kv 900):
+    # rows past kv=520 see their max jump by ~8*sqrt(d)*scale >> THR
+    with torch.no_grad():
+        k[:, :, 520, :] = 8.0
+        k[:, :, 900, :] = 4.0
+    with torch.no_grad():
+        o = flash_attention(q, k, v, causal=True)
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        q.float(), k.float(), v.float(), is_causal=True,
+        scale=1.0 / math.sqrt(d))
+    assert torch.isfinite(o.float()).all()
+    assert (o.float() - ref).abs().max() < 3e-2
