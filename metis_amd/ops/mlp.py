@@ -11,8 +11,14 @@ end-to-end here (see docs/KERNELS.md "Where library calls remain").
 GELU flavor is the tanh approximation (what the hipBLASLt epilogue
 implements); the CPU fallback matches it.
 
-Off by default: GPTBlock uses it only under METIS_FC1_EPILOGUE=1 until
-the numerics are validated on a GPU box (round-2 item).
+RETIRED on this stack (round-2 measurement): ROCm 7.2's hipBLASLt ships
+NO algorithm for HIPBLASLT_EPILOGUE_GELU_AUX_BIAS on gfx950 bf16 — the
+heuristic returns empty at every probed shape (4096^3, 8192x10240x2560,
+32768x10240x2560; gpurun_out/r2_profile epilogue probe). The default
+path (library GEMM + aten tanh-GELU, whose backward is a single fused
+gelu_backward kernel) therefore stays; METIS_FC1_EPILOGUE=1 raises at
+the first hipBLASLt call on current ROCm and is kept only so the route
+lights up automatically if a future hipBLASLt adds the algorithms.
 """
 
 from __future__ import annotations
