@@ -122,6 +122,9 @@ def profile_model(
     tp_group=None,
     seq_length: Optional[int] = None,
     recompute: bool = False,
+    emulate_tp: int = 0,
+    comm_bw_gbps: float = 130.0,
+    comm_alpha_us: float = 20.0,
 ) -> Optional[str]:
     """Profile one (tp, bs) point; returns the JSON path (rank 0).
 
@@ -131,11 +134,27 @@ def profile_model(
     how the planner prices recompute (measured, not modeled). Per-block
     times are then a uniform split of the block total: module backward
     hooks double-fire during checkpoint replay, and the blocks are
-    identical anyway."""
+    identical anyway.
+
+    ``emulate_tp=T`` (single-GPU boxes, where RCCL cannot host T ranks
+    on one device): build the model with the REAL per-rank tp=T shards
+    (compute and memory are measured exactly) and ADD the four
+    per-block TP all-reduces ([bs, s, h] bf16, the f/g operators) from
+    the alpha-beta comm model (2(T-1)/T * bytes / BW + alpha) to the
+    block times, fwd_bwd and the accumulation probes; the
+    LN/embedding-grad all-reduce keys are modeled the same way. Such
+    files carry ``tp_comm_modeled: true`` plus the constants used —
+    they are honest per-rank measurements with modeled collectives, to
+    be replaced by torchrun-measured profiles when a multi-GPU box is
+    available."""
     assert torch.cuda.is_available(), "profiler needs a GPU"
     if seq_length:
         import dataclasses
         spec = dataclasses.replace(spec, seq_length=seq_length)
+    if emulate_tp > 1:
+        assert tp == 1 and tp_group is None, (
+            "emulate_tp is for single-process runs")
+        tp = emulate_tp
     dev = torch.device("cuda", torch.cuda.current_device())
     if isinstance(spec, MoEModelSpec):
         model_cls = MoEModel
@@ -154,7 +173,13 @@ def profile_model(
 
     def batch():
         tokens = torch.randint(0, spec.vocab_size, (bs, spec.seq_length), device=dev)
-        return tokens, torch.roll(tokens, -1, 1)
+        labels = torch.roll(tokens, -1, 1)
+        if emulate_tp > 1:
+            # single-process emulation: the head is vocab-sharded but no
+            # group exists, so CE runs on the local shard — clamp labels
+            # into it (timing-faithful; the loss VALUE is not used)
+            labels = labels % (spec.vocab_size // emulate_tp)
+        return tokens, labels
 
     def one_iter(measure: bool):
         t_iter0 = time.perf_counter()
@@ -285,6 +310,32 @@ def profile_model(
     t4mb = sum(fwd_bwd_k(4) for _ in range(probe_iters)) / probe_iters
     opt.zero_grad()
 
+    extra_keys = None
+    if emulate_tp > 1:
+        def ar_ms(nbytes: float) -> float:
+            bw = comm_bw_gbps * 1024 * 1024   # clusterfile units: B/ms
+            return (comm_alpha_us / 1000.0
+                    + 2 * (emulate_tp - 1) / emulate_tp * nbytes / bw)
+
+        n_blocks = len(layer_ms) - 2
+        # 4 all-reduces of [bs, s, h] bf16 per block (f bwd x2, g fwd x2)
+        per_block = 4 * ar_ms(bs * spec.seq_length * spec.hidden_size * 2)
+        for i in range(1, len(layer_ms) - 1):
+            layer_ms[i] += per_block
+        acc["fwd_bwd"] += per_block * n_blocks
+        acc["total"] += per_block * n_blocks
+        t2mb += 2 * per_block * n_blocks
+        t4mb += 4 * per_block * n_blocks
+        # schema-prescribed grad all-reduce intervals (README.md:80-81)
+        acc["ln_ar"] = ar_ms(n_blocks * 4 * spec.hidden_size * 2)
+        acc["emb_ar"] = ar_ms(spec.vocab_size * spec.hidden_size * 2)
+        acc["total"] += acc["ln_ar"] + acc["emb_ar"]
+        extra_keys = {
+            "tp_comm_modeled": True,
+            "tp_comm_bw_GBps": comm_bw_gbps,
+            "tp_comm_alpha_us": comm_alpha_us,
+        }
+
     rank = dist.get_rank() if dist.is_initialized() else 0
     if rank != 0:
         return None
@@ -305,6 +356,7 @@ def profile_model(
         layer_memory_total_mb=layer_mem_mb,
         fwd_bwd_2mb_ms=t2mb,
         fwd_bwd_4mb_ms=t4mb,
+        extra_execution_keys=extra_keys,
     )
     return path
 
@@ -322,6 +374,13 @@ def main() -> None:
     p.add_argument("--recompute", action="store_true",
                    help="profile with per-block activation recomputation "
                         "(default out dir: profiles/mi355x_rc/<model>)")
+    p.add_argument("--emulate-tp", type=int, default=0,
+                   help="single-GPU tp emulation: measure the per-rank "
+                        "tp=T shards, model the collectives (see "
+                        "profile_model docstring)")
+    p.add_argument("--comm-bw", type=float, default=130.0,
+                   help="modeled all-reduce bus bandwidth for --emulate-tp")
+    p.add_argument("--comm-alpha-us", type=float, default=20.0)
     args = p.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -331,6 +390,7 @@ def main() -> None:
 
         ctx = init_parallel(dp=1, tp=world, pp=1)
         tp_group = ctx.tp_group
+        assert not args.emulate_tp, "--emulate-tp is single-process only"
 
     spec = MODEL_SPECS[args.model]
     base = "profiles/mi355x_rc" if args.recompute else "profiles/mi355x"
@@ -340,7 +400,8 @@ def main() -> None:
             spec, bs=bs, tp=world, device_type=args.device_type,
             out_dir=out_dir, warmup=args.warmup, iters=args.iters,
             tp_group=tp_group, seq_length=args.seq_length,
-            recompute=args.recompute,
+            recompute=args.recompute, emulate_tp=args.emulate_tp,
+            comm_bw_gbps=args.comm_bw, comm_alpha_us=args.comm_alpha_us,
         )
         if path:
             print(f"wrote {path}")

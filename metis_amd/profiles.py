@@ -249,6 +249,7 @@ class ProfileStore:
         fwd_bwd_1mb_ms: Optional[float] = None,
         fwd_bwd_2mb_ms: Optional[float] = None,
         fwd_bwd_4mb_ms: Optional[float] = None,
+        extra_execution_keys: Optional[dict] = None,
     ) -> None:
         doc = {
             "model": {
@@ -282,5 +283,7 @@ class ProfileStore:
             doc["execution_time"]["fwd_bwd_2mb_ms"] = fwd_bwd_2mb_ms
         if fwd_bwd_4mb_ms is not None:
             doc["execution_time"]["fwd_bwd_4mb_ms"] = fwd_bwd_4mb_ms
+        if extra_execution_keys:
+            doc["execution_time"].update(extra_execution_keys)
         with open(path, "w") as fh:
             json.dump(doc, fh, indent=2)
