@@ -309,7 +309,14 @@ class PlanRunner:
         stage instead of all B — the memory-bound pipeline schedule the
         reference's cost model prices but cannot run. Sends are isend (a
         steady-state fwd-send/bwd-send pair of neighboring stages would
-        deadlock with rendezvous sends); recvs stay blocking."""
+        deadlock with rendezvous sends); recvs stay blocking — and must:
+        p2p ops of one rank pair share a RCCL stream in issue order, and
+        1F1B interleaves the two directions, so a prefetched recv issued
+        ahead of this rank's next grad send can form a cycle with the
+        peer doing the same (recv-act(k+1) blocks send-grad(j) here,
+        while the peer's recv-grad(j) blocks its send-act(k+1)). GPipe's
+        phase structure has no such interleave, so only GPipe pre-posts
+        (see _step_pipeline)."""
         ctx = self.ctx
         nm = self.num_microbatches
         seq = self.spec.seq_length // (ctx.tp if self.sp else 1)
