@@ -185,7 +185,11 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
         //      just-issued write (measured -5% when adjacent).
         const int p4 = lane & 15;
         const bool diag = kv0 + KVBLK > qb;     // any masked column here?
-        float p[2][JSUB][4];   // [rb][j][r]: q rows k8*4+r, kv col
+        // ONE p array serves both row blocks: rb0's values are packed
+        // into the LDS P^T image (p_write) before qk_softmax(rb1)
+        // overwrites them — 32 fewer VGPRs at D=80 than a [2][...] p
+        // (which spilled 17 registers)
+        float p[JSUB][4];      // [j][r]: q rows k8*4+r, kv col
                                // j*16+col16 (C fragment layout)
         bool rb_active[2];
 
@@ -224,7 +228,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
                         const int kvcol = kv0 + j * 16 + col16;
                         sv = (kvcol > qrow) ? -1e30f : sv;
                     }
-                    p[rb][j][r] = sv;
+                    p[j][r] = sv;
                 }
             }
             __builtin_amdgcn_s_setprio(0);
@@ -235,9 +239,9 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
             bool need = false;
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                float tm = p[rb][0][r];
+                float tm = p[0][r];
                 #pragma unroll
-                for (int j = 1; j < JSUB; ++j) tm = fmaxf(tm, p[rb][j][r]);
+                for (int j = 1; j < JSUB; ++j) tm = fmaxf(tm, p[j][r]);
                 #pragma unroll
                 for (int off = 8; off > 0; off >>= 1)
                     tm = fmaxf(tm, __shfl_xor(tm, off, 16));
@@ -261,8 +265,8 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
                 float row_sum = 0.f;
                 #pragma unroll
                 for (int j = 0; j < JSUB; ++j) {
-                    p[rb][j][r] = exp2f(p[rb][j][r] - m_run[rb][r]);
-                    row_sum += p[rb][j][r];
+                    p[j][r] = exp2f(p[j][r] - m_run[rb][r]);
+                    row_sum += p[j][r];
                 }
                 #pragma unroll
                 for (int off = 8; off > 0; off >>= 1)
@@ -279,7 +283,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
                 short4v pk;
                 #pragma unroll
                 for (int r = 0; r < 4; ++r)
-                    pk[r] = float_to_bf16_bits(p[rb][j][r]);
+                    pk[r] = float_to_bf16_bits(p[j][r]);
                 *reinterpret_cast<short4v*>(
                     Pw + (j * 16 + col16) * PROW + k8 * 4) = pk;
             }

@@ -320,7 +320,8 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     // VGPRs on top of the dK+dV accumulators, so dO stages direct there.
     constexpr int chunks = QT * D / 8;
     constexpr int per_thread = (chunks + THREADS - 1) / THREADS;
-    constexpr bool STAGE_DO = (D <= 96 && QT <= 64);
+    constexpr bool STAGE_DO = (D <= 96);   // prefetch dO too (un-staged
+                                           // dO loads cost -18% at QT=128)
     bf16x8 q_stage[per_thread];
     bf16x8 do_stage[STAGE_DO ? per_thread : 1];
     auto issue_loads = [&](int q0) {
