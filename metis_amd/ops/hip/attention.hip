@@ -27,18 +27,7 @@ typedef short short4v __attribute__((ext_vector_type(4)));
 
 constexpr int THREADS = 256;
 constexpr int QBLK = 128;     // q rows per block (2 x 16 per wave)
-constexpr int PROW = 20;      // P^T LDS row length (16 q + 4 pad shorts):
-                              // 40-B row stride keeps the packed b64 P
-                              // stores conflict-free (10*i mod 32 distinct
-                              // over a 16-lane group) and tr16 reads
-                              // 8-B aligned
-// defer-max threshold (T13, log2 domain): skip the O/l rescale while the
-// running max grows by < THR — P is then bounded by 2^THR instead of 1,
-// which the fp32 l/O accumulators absorb (P itself is bf16: ~3x looser
-// abs error on O, still ~1e-2-class for unit-variance inputs)
-constexpr float RESCALE_THR = 8.0f;
-constexpr float LOG2E = 1.4426950408889634f;
-constexpr float LN2 = 0.6931471805599453f;
+constexpr int VPAD = 8;       // Vt row padding (bf16 elements)
 
 // KV tile width: 128 for D <= 80 (the K/V^T/P LDS images still fit two
 // blocks per CU), 64 above. Wider tiles amortize the two barriers per
@@ -65,6 +54,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
     constexpr int dchunks = (D + 31) / 32;   // 32-wide K-dim chunks of D
     constexpr int djtiles = D / 16;          // 16-wide output column tiles
     constexpr int KSLOT = D / 8 + 1;         // K LDS slots per row (padded)
+    constexpr int VROW = KVBLK + VPAD;       // Vt LDS row length
 
     const int qtile = blockIdx.x % (S / QBLK);
     const int head = (blockIdx.x / (S / QBLK)) % H;
@@ -83,14 +73,10 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
     // numerically — never store element-wise through the struct type).
     // V stays in its NATURAL [kv][D] image: PV B-fragments read it with
     // ds_read_b64_tr_b16 (hardware transpose), so no scalar transpose
-    // staging exists anywhere in this kernel. P is staged TRANSPOSED
-    // ([kv][16 q], one image per wave, reused across the wave's two row
-    // blocks): the C-fragment's 4 q values per lane are one packed
-    // ds_write_b64, and the PV A-fragment comes back via the same tr16
-    // hardware-transpose read the V path uses — no scalar LDS traffic.
+    // staging exists anywhere in this kernel.
     short* Ks = reinterpret_cast<short*>(smem);                 // [KVBLK][KSLOT*8]
     short* Vs = Ks + KVBLK * KSLOT * 8;                         // [KVBLK][KSLOT*8]
-    short* Pw = Vs + KVBLK * KSLOT * 8 + wave * KVBLK * PROW;   // [KVBLK][PROW]
+    short* Pw = Vs + KVBLK * KSLOT * 8 + wave * 2 * 16 * VROW;  // [2][16][VROW]
 
     // ---- preload Q fragments for both row blocks ------------------------
     bf16x8 q_frag[2][dchunks];
@@ -175,26 +161,15 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
         if (kv0 + KVBLK < kv_end) issue_loads(kv0 + KVBLK);
         __syncthreads();
 
-        // ---- per row block: S = scale * Q K^T, online softmax, P^T ->
-        //      LDS (packed), O += P V. Phase order interleaves the two
-        //      row blocks — [QK+SM 0][W0][QK+SM 1][PV0][W1][PV1] — so a
-        //      P^T write burst is separated from its own tr16 re-reads
-        //      by ~a full QK+softmax (or PV) phase: the single per-wave
-        //      P^T image is written for rb1 only after rb0's PV has
-        //      consumed it (per-wave DS ordering), and no read chases a
-        //      just-issued write (measured -5% when adjacent).
-        const int p4 = lane & 15;
-        const bool diag = kv0 + KVBLK > qb;     // any masked column here?
-        // ONE p array serves both row blocks: rb0's values are packed
-        // into the LDS P^T image (p_write) before qk_softmax(rb1)
-        // overwrites them — 32 fewer VGPRs at D=80 than a [2][...] p
-        // (which spilled 17 registers)
-        float p[JSUB][4];      // [j][r]: q rows k8*4+r, kv col
-                               // j*16+col16 (C fragment layout)
+        // ---- per row block: S = scale * Q K^T, online softmax, P -> LDS -
         bool rb_active[2];
-
-        auto qk_softmax = [&](int rb) {
+        float p[2][JSUB][4];   // [rb][j][r]
+        #pragma unroll
+        for (int rb = 0; rb < 2; ++rb) {
             const int q0 = qb + rbid[rb] * 16;
+            rb_active[rb] = kv0 <= q0 + 15;
+            if (!rb_active[rb]) continue;
+
             __builtin_amdgcn_s_setprio(1);   // T5: favor the MFMA cluster
             #pragma unroll
             for (int j = 0; j < JSUB; ++j) {    // 16-col subtiles
@@ -216,99 +191,71 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
                 }
                 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
-                    // log2 domain: v_exp_f32 natively computes 2^x, so
-                    // folding log2(e) into the scale deletes one multiply
-                    // per element of every later exp
-                    float sv = s_acc[r] * (scale * LOG2E);
-                    if (diag) {
-                        // causal select only on diagonal tiles (interior
-                        // tiles have no masked columns — 32 fewer VALU
-                        // selects per row block there)
-                        const int qrow = q0 + k8 * 4 + r;
-                        const int kvcol = kv0 + j * 16 + col16;
-                        sv = (kvcol > qrow) ? -1e30f : sv;
-                    }
-                    p[j][r] = sv;
+                    const int qrow = q0 + k8 * 4 + r;
+                    const int kvcol = kv0 + j * 16 + col16;
+                    float sv = s_acc[r] * scale;
+                    p[rb][j][r] = (kvcol > qrow) ? -1e30f : sv;
                 }
             }
             __builtin_amdgcn_s_setprio(0);
 
-            // T13 defer-max: only rescale O/l when some row's max grew
-            // past THR (in log2 units); P stays bounded by 2^THR
-            float tile_max[4];
-            bool need = false;
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                float tm = p[0][r];
+                float tile_max = -1e30f;
                 #pragma unroll
-                for (int j = 1; j < JSUB; ++j) tm = fmaxf(tm, p[j][r]);
+                for (int j = 0; j < JSUB; ++j)
+                    tile_max = fmaxf(tile_max, p[rb][j][r]);
                 #pragma unroll
                 for (int off = 8; off > 0; off >>= 1)
-                    tm = fmaxf(tm, __shfl_xor(tm, off, 16));
-                tile_max[r] = tm;
-                need = need || (tm > m_run[rb][r] + RESCALE_THR);
-            }
-            if (__any(need)) {
-                #pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    const float m_new = fmaxf(m_run[rb][r], tile_max[r]);
-                    const float alpha = exp2f(m_run[rb][r] - m_new);
-                    m_run[rb][r] = m_new;
-                    l_run[rb][r] *= alpha;
-                    #pragma unroll
-                    for (int jd = 0; jd < djtiles; ++jd)
-                        o_acc[rb][jd][r] *= alpha;
-                }
-            }
-            #pragma unroll
-            for (int r = 0; r < 4; ++r) {
+                    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, off, 16));
+
+                const float m_new = fmaxf(m_run[rb][r], tile_max);
+                const float alpha = __expf(m_run[rb][r] - m_new);
                 float row_sum = 0.f;
                 #pragma unroll
                 for (int j = 0; j < JSUB; ++j) {
-                    p[j][r] = exp2f(p[j][r] - m_run[rb][r]);
-                    row_sum += p[j][r];
+                    p[rb][j][r] = __expf(p[rb][j][r] - m_new);
+                    row_sum += p[rb][j][r];
                 }
                 #pragma unroll
                 for (int off = 8; off > 0; off >>= 1)
                     row_sum += __shfl_xor(row_sum, off, 16);
-                l_run[rb][r] += row_sum;
-            }
-        };
 
-        // P^T image: row kv = j*16+col16, cols q = k8*4..+3 — the
-        // lane's 4 C-fragment values are one packed 8-B store
-        auto p_write = [&](int rb) {
+                l_run[rb][r] = l_run[rb][r] * alpha + row_sum;
+                m_run[rb][r] = m_new;
+                #pragma unroll
+                for (int jd = 0; jd < djtiles; ++jd)
+                    o_acc[rb][jd][r] *= alpha;
+            }
+
+            short* Prb = Pw + rb * 16 * VROW;
             #pragma unroll
-            for (int j = 0; j < JSUB; ++j) {
-                short4v pk;
+            for (int j = 0; j < JSUB; ++j)
                 #pragma unroll
                 for (int r = 0; r < 4; ++r)
-                    pk[r] = float_to_bf16_bits(p[j][r]);
-                *reinterpret_cast<short4v*>(
-                    Pw + (j * 16 + col16) * PROW + k8 * 4) = pk;
-            }
-            asm volatile("" ::: "memory");   // per-wave DS ordering
-        };
+                    Prb[(k8 * 4 + r) * VROW + j * 16 + col16] =
+                        float_to_bf16_bits(p[rb][j][r]);
+        }
 
-        // O += P @ V: both A (P^T image) and B (V natural image)
-        // fragments via tr16 hardware-transpose reads — each 16-lane
-        // group loads a 4(kv)x16 block, lane l&15 receiving its
-        // column's 4 kv values.
-        auto pv = [&](int rb) {
+        // P is per-wave private: DS ops of one wave complete in order, so
+        // a compiler-level fence (no barrier) suffices to keep the vector
+        // re-read below the scalar writes above.
+        asm volatile("" ::: "memory");
+
+        // ---- O += P @ V -------------------------------------------------
+        #pragma unroll
+        for (int rb = 0; rb < 2; ++rb) {
+            if (!rb_active[rb]) continue;
+            const short* Prb = Pw + rb * 16 * VROW;
+            // V B-fragments via hardware transpose read: each 16-lane
+            // group cooperatively loads one 4(kv)x16(d) block — lane
+            // l&15 receives its d-column's 4 kv values. Per fragment,
+            // two tr16 reads cover the 8 kv rows of this lane group.
+            const int p4 = lane & 15;     // piece index within the group
             #pragma unroll
             for (int ks = 0; ks < KVBLK / 32; ++ks) {   // 32-wide kv chunks
-                bf16x8 p_frag;
-                #pragma unroll
-                for (int r = 0; r < 2; ++r) {
-                    const int kvrow = ks * 32 + k8 * 8 + 4 * r + (p4 >> 2);
-                    short4v t = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-                        (__attribute__((address_space(3))) short4v*)(
-                            Pw + kvrow * PROW + (p4 & 3) * 4));
-                    p_frag[r * 4 + 0] = t[0];
-                    p_frag[r * 4 + 1] = t[1];
-                    p_frag[r * 4 + 2] = t[2];
-                    p_frag[r * 4 + 3] = t[3];
-                }
+                bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(
+                    Prb + col16 * VROW + ks * 32 + k8 * 8);
                 #pragma unroll
                 for (int jd = 0; jd < djtiles; ++jd) {
                     bf16x8 v_frag;
@@ -328,19 +275,6 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
                         p_frag, v_frag, o_acc[rb][jd], 0, 0, 0);
                 }
             }
-        };
-
-        rb_active[0] = kv0 <= qb + rbid[0] * 16 + 15;
-        rb_active[1] = kv0 <= qb + rbid[1] * 16 + 15;
-        if (rb_active[0]) {
-            qk_softmax(0);
-            p_write(0);
-        }
-        if (rb_active[1]) qk_softmax(1);
-        if (rb_active[0]) pv(0);
-        if (rb_active[1]) {
-            p_write(1);
-            pv(1);
         }
         __syncthreads();   // K/Vt/P reused next tile
     }
@@ -357,10 +291,8 @@ __global__ __launch_bounds__(THREADS, 2) void attn_fwd_kernel(
                 O[q_base + (long)qrow * D + jd * 16 + col16] =
                     __float2bfloat16(o_acc[rb][jd][r] * inv_l);
             if (col16 == 0)
-                // m_run lives in the log2 domain (see QK^T scale fold);
-                // LSE stays natural-log for the backward pass
                 LSE[((long)batch * H + head) * S + qrow] =
-                    m_run[rb][r] * LN2 + __logf(l_run[rb][r]);
+                    m_run[rb][r] + __logf(l_run[rb][r]);
         }
     }
 }
@@ -387,7 +319,7 @@ std::vector<torch::Tensor> attn_fwd(
         do {                                                                  \
             const int kvb = kvblk_for<DD>();                                  \
             const int lds = (2 * kvb * (DD / 8 + 1) * 8                       \
-                             + 4 * kvb * PROW) * 2;                           \
+                             + 4 * 2 * 16 * (kvb + VPAD)) * 2;                \
             hipLaunchKernelGGL(attn_fwd_kernel<DD>, dim3(grid),               \
                 dim3(THREADS), lds, stream,                                   \
                 reinterpret_cast<const bf16*>(qc.data_ptr()),                 \

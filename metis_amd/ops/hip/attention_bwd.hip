@@ -270,11 +270,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
 
     constexpr int dchunks = (D + 31) / 32;
     constexpr int djtiles = D / 16;
-    // q-tile width: 128 at D <= 80 (the kernel is barrier/wait-bound —
-    // PMC r2d: 57-64% SQ_WAIT_ANY at QT=64 — and the doubled Q/dO
-    // images still fit 2 blocks/CU below D=96), 64 above
-    constexpr int QT = D <= 80 ? 128 : 64;
-    constexpr int VROW = QT + VPAD;
+    constexpr int VROW = CTILE + VPAD;
     constexpr int KSLOT = D / 8 + 1;
     constexpr int KVB = 64;            // kv rows per block (16 per wave)
 
@@ -294,10 +290,10 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     // Qt/dOt copies (and their LDS staging writes) are gone; the freed
     // LDS restores 2 blocks/CU at every D.
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    short* Qs = reinterpret_cast<short*>(smem);                    // [QT][KSLOT*8]
-    short* dOs = Qs + QT * KSLOT * 8;                              // [QT][KSLOT*8]
-    short* Sw = dOs + QT * KSLOT * 8 + wave * 16 * VROW;           // [16][VROW]
-    short* Pw = dOs + QT * KSLOT * 8 + 4 * 16 * VROW
+    short* Qs = reinterpret_cast<short*>(smem);                    // [CTILE][KSLOT*8]
+    short* dOs = Qs + CTILE * KSLOT * 8;                           // [CTILE][KSLOT*8]
+    short* Sw = dOs + CTILE * KSLOT * 8 + wave * 16 * VROW;        // [16][VROW]
+    short* Pw = dOs + CTILE * KSLOT * 8 + 4 * 16 * VROW
                 + wave * 16 * VROW;                                // [16][VROW]
 
     // K and V fragments for this wave's rows (A layout, m = col16)
@@ -318,10 +314,9 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     // T14 staged Q and dO (each register set feeds both the natural and
     // transposed LDS image); at D=128 the dO set would spill past 256
     // VGPRs on top of the dK+dV accumulators, so dO stages direct there.
-    constexpr int chunks = QT * D / 8;
+    constexpr int chunks = CTILE * D / 8;
     constexpr int per_thread = (chunks + THREADS - 1) / THREADS;
-    constexpr bool STAGE_DO = (D <= 96);   // prefetch dO too (un-staged
-                                           // dO loads cost -18% at QT=128)
+    constexpr bool STAGE_DO = (D <= 96);
     bf16x8 q_stage[per_thread];
     bf16x8 do_stage[STAGE_DO ? per_thread : 1];
     auto issue_loads = [&](int q0) {
@@ -330,10 +325,10 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
             const int c = threadIdx.x + u * THREADS;
             if (c >= chunks) break;
             q_stage[u] = *reinterpret_cast<const bf16x8*>(
-                Q + q_base + (long)(q0 + c % QT) * D + (c / QT) * 8);
+                Q + q_base + (long)(q0 + c % CTILE) * D + (c / CTILE) * 8);
             if constexpr (STAGE_DO)
                 do_stage[u] = *reinterpret_cast<const bf16x8*>(
-                    dO + q_base + (long)(q0 + c % QT) * D + (c / QT) * 8);
+                    dO + q_base + (long)(q0 + c % CTILE) * D + (c / CTILE) * 8);
         }
     };
     auto write_stage = [&](int q0) {
@@ -341,8 +336,8 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
         for (int u = 0; u < per_thread; ++u) {
             const int c = threadIdx.x + u * THREADS;
             if (c >= chunks) break;
-            const int row = c % QT;
-            const int d0 = (c / QT) * 8;
+            const int row = c % CTILE;
+            const int d0 = (c / CTILE) * 8;
             *reinterpret_cast<bf16x8*>(Qs + row * KSLOT * 8 + d0) = q_stage[u];
             bf16x8 dov;
             if constexpr (STAGE_DO)
@@ -354,77 +349,66 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
         }
     };
 
-    // causal start, rounded DOWN to a QT boundary so the stage walk
-    // stays QT-aligned (the sub-QT inactive half is masked per wave)
-    const int q_start = (kvtile * KVB) / QT * QT;
+    const int q_start = kvtile * KVB;   // causal: from the block's kv start
     issue_loads(q_start);
-    for (int q0 = q_start; q0 < S; q0 += QT) {
+    for (int q0 = q_start; q0 < S; q0 += CTILE) {
         write_stage(q0);
-        if (q0 + QT < S) issue_loads(q0 + QT);
+        if (q0 + CTILE < S) issue_loads(q0 + CTILE);
         __syncthreads();
 
-        const bool active = q0 + QT - 1 >= kv0;
+        const bool active = q0 + CTILE - 1 >= kv0;
         if (active) {
-            // 64-q-column halves: keeps the st/dpt accumulator arrays
-            // at 4 fragments (QT=128 at full width would spill) and
-            // puts a full half's MFMA+exp phase between a Sw/Pw write
-            // burst and the dk/dv phase's reads
+            __builtin_amdgcn_s_setprio(1);
+            floatx4 st_acc[4], dpt_acc[4];
             #pragma unroll
-            for (int h = 0; h < QT / 64; ++h) {
-                __builtin_amdgcn_s_setprio(1);
-                floatx4 st_acc[4], dpt_acc[4];
+            for (int j = 0; j < 4; ++j) {
+                st_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
+                dpt_acc[j] = floatx4{0.f, 0.f, 0.f, 0.f};
+                const int qrow = j * 16 + col16;
                 #pragma unroll
-                for (int jj = 0; jj < 4; ++jj) {
-                    const int j = h * 4 + jj;
-                    st_acc[jj] = floatx4{0.f, 0.f, 0.f, 0.f};
-                    dpt_acc[jj] = floatx4{0.f, 0.f, 0.f, 0.f};
-                    const int qrow = j * 16 + col16;
-                    #pragma unroll
-                    for (int c = 0; c < dchunks; ++c) {
-                        const int d0 = c * 32 + k8 * 8;
-                        bf16x8 qf, dof;
-                        if (d0 < D) {
-                            qf = *reinterpret_cast<const bf16x8*>(
-                                Qs + qrow * KSLOT * 8 + d0);
-                            dof = *reinterpret_cast<const bf16x8*>(
-                                dOs + qrow * KSLOT * 8 + d0);
-                        } else {
-                            #pragma unroll
-                            for (int i = 0; i < 8; ++i) { qf[i] = 0; dof[i] = 0; }
-                        }
-                        st_acc[jj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            k_frag[c], qf, st_acc[jj], 0, 0, 0);
-                        dpt_acc[jj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            v_frag[c], dof, dpt_acc[jj], 0, 0, 0);
+                for (int c = 0; c < dchunks; ++c) {
+                    const int d0 = c * 32 + k8 * 8;
+                    bf16x8 qf, dof;
+                    if (d0 < D) {
+                        qf = *reinterpret_cast<const bf16x8*>(
+                            Qs + qrow * KSLOT * 8 + d0);
+                        dof = *reinterpret_cast<const bf16x8*>(
+                            dOs + qrow * KSLOT * 8 + d0);
+                    } else {
+                        #pragma unroll
+                        for (int i = 0; i < 8; ++i) { qf[i] = 0; dof[i] = 0; }
                     }
+                    st_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        k_frag[c], qf, st_acc[j], 0, 0, 0);
+                    dpt_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        v_frag[c], dof, dpt_acc[j], 0, 0, 0);
                 }
-                __builtin_amdgcn_s_setprio(0);
-
-                #pragma unroll
-                for (int jj = 0; jj < 4; ++jj)
-                    #pragma unroll
-                    for (int r = 0; r < 4; ++r) {
-                        const int j = h * 4 + jj;
-                        const int kvrow = kv0 + k8 * 4 + r;
-                        const int qcol = q0 + j * 16 + col16;
-                        float p = (qcol < kvrow)
-                                      ? 0.f
-                                      : __expf(st_acc[jj][r] * scale
-                                               - LSE[row_base + qcol]);
-                        float ds = scale * p
-                                   * (dpt_acc[jj][r] - Delta[row_base + qcol]);
-                        Sw[(k8 * 4 + r) * VROW + j * 16 + col16] =
-                            float_to_bf16_bits(ds);
-                        Pw[(k8 * 4 + r) * VROW + j * 16 + col16] =
-                            float_to_bf16_bits(p);
-                    }
             }
+            __builtin_amdgcn_s_setprio(0);
+
+            #pragma unroll
+            for (int j = 0; j < 4; ++j)
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int kvrow = kv0 + k8 * 4 + r;
+                    const int qcol = q0 + j * 16 + col16;
+                    float p = (qcol < kvrow)
+                                  ? 0.f
+                                  : __expf(st_acc[j][r] * scale
+                                           - LSE[row_base + qcol]);
+                    float ds = scale * p
+                               * (dpt_acc[j][r] - Delta[row_base + qcol]);
+                    Sw[(k8 * 4 + r) * VROW + j * 16 + col16] =
+                        float_to_bf16_bits(ds);
+                    Pw[(k8 * 4 + r) * VROW + j * 16 + col16] =
+                        float_to_bf16_bits(p);
+                }
 
             asm volatile("" ::: "memory");   // wave-local publish
 
             const int p4 = lane & 15;
             #pragma unroll
-            for (int ks = 0; ks < QT / 32; ++ks) {
+            for (int ks = 0; ks < 2; ++ks) {
                 bf16x8 dst_frag = *reinterpret_cast<const bf16x8*>(
                     Sw + col16 * VROW + ks * 32 + k8 * 8);
                 bf16x8 pt_frag = *reinterpret_cast<const bf16x8*>(
@@ -485,10 +469,8 @@ std::vector<torch::Tensor> attn_bwd(
             const int vrow = CTILE + VPAD;                                    \
             const int lds_dq = (2 * CTILE * (DD / 8 + 1) * 8                  \
                                 + DD * vrow + 4 * 2 * 16 * vrow) * 2;         \
-            const int qt = DD <= 80 ? 128 : 64;                               \
-            const int vrow_kv = qt + VPAD;                                    \
-            const int lds_dkv = (2 * qt * (DD / 8 + 1) * 8                    \
-                                 + 2 * 4 * 16 * vrow_kv) * 2;                 \
+            const int lds_dkv = (2 * CTILE * (DD / 8 + 1) * 8                 \
+                                 + 2 * 4 * 16 * vrow) * 2;                    \
             hipLaunchKernelGGL(attn_bwd_dq_kernel<DD>, dim3(grid),            \
                 dim3(THREADS), lds_dq, stream,                                \
                 reinterpret_cast<const bf16*>(qc.data_ptr()),                 \
