@@ -42,7 +42,7 @@ def main():
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     tps = args.batch * args.new / dt
-    flag = os.environ.get("METIS_DECODE_KERNEL", "0")
+    flag = os.environ.get("METIS_DECODE_KERNEL", "1(default)")
     print(f"decode {args.model} b{args.batch} p{args.prompt} n{args.new} "
           f"METIS_DECODE_KERNEL={flag}: {dt*1e3:.1f} ms, {tps:.0f} tok/s")
     assert out.shape == (args.batch, args.prompt + args.new)
