@@ -252,11 +252,14 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
 }
 
 // --------------------------------------------------------------- dKV ----
-// One 16-row kv block per wave (64 per block): dkv's causal skip is at
-// most one q tile per wave, so row-block pairing buys nothing and the
-// freed registers/LDS go to staged natural Q/dO tiles instead.
+// One 16-row kv block per wave, EIGHT waves per block (128 kv rows):
+// the whole block shares one staged Q/dO stream, so doubling the waves
+// per block halves the number of Q/dO sweeps over the sequence (the
+// kernel is wait-bound - 57-64% SQ_WAIT_ANY in the r2 PMC run - and
+// the staging stream is the thing being waited on). Occupancy is
+// unchanged: one 8-wave block/CU at VGPR>128 = the old 2x4-wave.
 template <int D>
-__global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
+__global__ __launch_bounds__(2 * THREADS, 2) void attn_bwd_dkv_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, const bf16* __restrict__ dO,
     const float* __restrict__ LSE, const float* __restrict__ Delta,
@@ -272,7 +275,8 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     constexpr int djtiles = D / 16;
     constexpr int VROW = CTILE + VPAD;
     constexpr int KSLOT = D / 8 + 1;
-    constexpr int KVB = 64;            // kv rows per block (16 per wave)
+    constexpr int KVB = 128;           // kv rows per block (16 per wave)
+    constexpr int NW = KVB / 16;       // 8 waves
 
     const int kvtile = blockIdx.x % (S / KVB);
     const int head = (blockIdx.x / (S / KVB)) % H;
@@ -293,7 +297,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     short* Qs = reinterpret_cast<short*>(smem);                    // [CTILE][KSLOT*8]
     short* dOs = Qs + CTILE * KSLOT * 8;                           // [CTILE][KSLOT*8]
     short* Sw = dOs + CTILE * KSLOT * 8 + wave * 16 * VROW;        // [16][VROW]
-    short* Pw = dOs + CTILE * KSLOT * 8 + 4 * 16 * VROW
+    short* Pw = dOs + CTILE * KSLOT * 8 + NW * 16 * VROW
                 + wave * 16 * VROW;                                // [16][VROW]
 
     // K and V fragments for this wave's rows (A layout, m = col16)
@@ -314,15 +318,16 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     // T14 staged Q and dO (each register set feeds both the natural and
     // transposed LDS image); at D=128 the dO set would spill past 256
     // VGPRs on top of the dK+dV accumulators, so dO stages direct there.
+    constexpr int NT = 2 * THREADS;     // 512 staging threads
     constexpr int chunks = CTILE * D / 8;
-    constexpr int per_thread = (chunks + THREADS - 1) / THREADS;
+    constexpr int per_thread = (chunks + NT - 1) / NT;
     constexpr bool STAGE_DO = (D <= 96);
     bf16x8 q_stage[per_thread];
     bf16x8 do_stage[STAGE_DO ? per_thread : 1];
     auto issue_loads = [&](int q0) {
         #pragma unroll
         for (int u = 0; u < per_thread; ++u) {
-            const int c = threadIdx.x + u * THREADS;
+            const int c = threadIdx.x + u * NT;
             if (c >= chunks) break;
             q_stage[u] = *reinterpret_cast<const bf16x8*>(
                 Q + q_base + (long)(q0 + c % CTILE) * D + (c / CTILE) * 8);
@@ -334,7 +339,7 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dkv_kernel(
     auto write_stage = [&](int q0) {
         #pragma unroll
         for (int u = 0; u < per_thread; ++u) {
-            const int c = threadIdx.x + u * THREADS;
+            const int c = threadIdx.x + u * NT;
             if (c >= chunks) break;
             const int row = c % CTILE;
             const int d0 = (c / CTILE) * 8;
@@ -462,7 +467,7 @@ std::vector<torch::Tensor> attn_bwd(
 
     auto stream = c10::hip::getCurrentHIPStream().stream();
     const int grid = (int)(B * H * (S / RBLK));
-    const int grid_dkv = (int)(B * H * (S / 64));
+    const int grid_dkv = (int)(B * H * (S / 128));   // 8-wave blocks
 
     #define LAUNCH_BWD(DD)                                                    \
         do {                                                                  \
@@ -470,7 +475,7 @@ std::vector<torch::Tensor> attn_bwd(
             const int lds_dq = (2 * CTILE * (DD / 8 + 1) * 8                  \
                                 + DD * vrow + 4 * 2 * 16 * vrow) * 2;         \
             const int lds_dkv = (2 * CTILE * (DD / 8 + 1) * 8                 \
-                                 + 2 * 4 * 16 * vrow) * 2;                    \
+                                 + 2 * 8 * 16 * vrow) * 2;                    \
             hipLaunchKernelGGL(attn_bwd_dq_kernel<DD>, dim3(grid),            \
                 dim3(THREADS), lds_dq, stream,                                \
                 reinterpret_cast<const bf16*>(qc.data_ptr()),                 \
@@ -481,7 +486,7 @@ std::vector<torch::Tensor> attn_bwd(
                 reinterpret_cast<bf16*>(dq.data_ptr()),                       \
                 (int)B, (int)H, (int)Hkv, (int)S, (float)scale);              \
             hipLaunchKernelGGL(attn_bwd_dkv_kernel<DD>, dim3(grid_dkv),       \
-                dim3(THREADS), lds_dkv, stream,                               \
+                dim3(2 * THREADS), lds_dkv, stream,                           \
                 reinterpret_cast<const bf16*>(qc.data_ptr()),                 \
                 reinterpret_cast<const bf16*>(kc.data_ptr()),                 \
                 reinterpret_cast<const bf16*>(vc.data_ptr()),                 \
