@@ -252,14 +252,16 @@ __global__ __launch_bounds__(THREADS, 2) void attn_bwd_dq_kernel(
 }
 
 // --------------------------------------------------------------- dKV ----
-// One 16-row kv block per wave, EIGHT waves per block (128 kv rows):
-// the whole block shares one staged Q/dO stream, so doubling the waves
-// per block halves the number of Q/dO sweeps over the sequence (the
-// kernel is wait-bound - 57-64% SQ_WAIT_ANY in the r2 PMC run - and
-// the staging stream is the thing being waited on). Occupancy is
-// unchanged: one 8-wave block/CU at VGPR>128 = the old 2x4-wave.
+// One 16-row kv block per wave. At D=64 the block is EIGHT waves (128
+// kv rows): all waves share one staged Q/dO stream, halving the number
+// of Q/dO sweeps over the sequence (the kernel is wait-bound - 57-64%
+// SQ_WAIT_ANY in the r2 PMC run). Measured: +15% at D=64; at D=80 the
+// same change was -8.5% and at D=128/S=4096 -9.5% (deeper per-wave
+// VGPR footprints lose the second independent block's latency
+// cover), so those stay 4-wave x 64 rows.
 template <int D>
-__global__ __launch_bounds__(2 * THREADS, 2) void attn_bwd_dkv_kernel(
+__global__ __launch_bounds__(D == 64 ? 2 * THREADS : THREADS, 2)
+void attn_bwd_dkv_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, const bf16* __restrict__ dO,
     const float* __restrict__ LSE, const float* __restrict__ Delta,
@@ -275,8 +277,8 @@ __global__ __launch_bounds__(2 * THREADS, 2) void attn_bwd_dkv_kernel(
     constexpr int djtiles = D / 16;
     constexpr int VROW = CTILE + VPAD;
     constexpr int KSLOT = D / 8 + 1;
-    constexpr int KVB = 128;           // kv rows per block (16 per wave)
-    constexpr int NW = KVB / 16;       // 8 waves
+    constexpr int KVB = D == 64 ? 128 : 64;  // kv rows per block
+    constexpr int NW = KVB / 16;             // waves per block
 
     const int kvtile = blockIdx.x % (S / KVB);
     const int head = (blockIdx.x / (S / KVB)) % H;
@@ -318,7 +320,7 @@ __global__ __launch_bounds__(2 * THREADS, 2) void attn_bwd_dkv_kernel(
     // T14 staged Q and dO (each register set feeds both the natural and
     // transposed LDS image); at D=128 the dO set would spill past 256
     // VGPRs on top of the dK+dV accumulators, so dO stages direct there.
-    constexpr int NT = 2 * THREADS;     // 512 staging threads
+    constexpr int NT = NW * 64;         // staging threads
     constexpr int chunks = CTILE * D / 8;
     constexpr int per_thread = (chunks + NT - 1) / NT;
     constexpr bool STAGE_DO = (D <= 96);
@@ -467,15 +469,16 @@ std::vector<torch::Tensor> attn_bwd(
 
     auto stream = c10::hip::getCurrentHIPStream().stream();
     const int grid = (int)(B * H * (S / RBLK));
-    const int grid_dkv = (int)(B * H * (S / 128));   // 8-wave blocks
+    // dkv grid: one block per KVB kv rows (KVB is 128 at D=64, 64 else)
 
     #define LAUNCH_BWD(DD)                                                    \
         do {                                                                  \
             const int vrow = CTILE + VPAD;                                    \
             const int lds_dq = (2 * CTILE * (DD / 8 + 1) * 8                  \
                                 + DD * vrow + 4 * 2 * 16 * vrow) * 2;         \
+            const int nw_dkv = (DD == 64) ? 8 : 4;                            \
             const int lds_dkv = (2 * CTILE * (DD / 8 + 1) * 8                 \
-                                 + 2 * 8 * 16 * vrow) * 2;                    \
+                                 + 2 * nw_dkv * 16 * vrow) * 2;               \
             hipLaunchKernelGGL(attn_bwd_dq_kernel<DD>, dim3(grid),            \
                 dim3(THREADS), lds_dq, stream,                                \
                 reinterpret_cast<const bf16*>(qc.data_ptr()),                 \
@@ -485,8 +488,10 @@ std::vector<torch::Tensor> attn_bwd(
                 lsec.data_ptr<float>(), dc.data_ptr<float>(),                 \
                 reinterpret_cast<bf16*>(dq.data_ptr()),                       \
                 (int)B, (int)H, (int)Hkv, (int)S, (float)scale);              \
+            const int kvb_dkv = (DD == 64) ? 128 : 64;                        \
+            const int grid_dkv = (int)(B * H * (S / kvb_dkv));                \
             hipLaunchKernelGGL(attn_bwd_dkv_kernel<DD>, dim3(grid_dkv),       \
-                dim3(2 * THREADS), lds_dkv, stream,                           \
+                dim3((DD == 64) ? 2 * THREADS : THREADS), lds_dkv, stream,    \
                 reinterpret_cast<const bf16*>(qc.data_ptr()),                 \
                 reinterpret_cast<const bf16*>(kc.data_ptr()),                 \
                 reinterpret_cast<const bf16*>(vc.data_ptr()),                 \
