@@ -10,9 +10,12 @@ random-init model for smoke testing:
                   "temperature": 0.8, "top_k": 40}
   -> {"tokens": [[prompt + continuation]]}
 
-v1 scope: single process (tp=pp=1), token-id interface (tokenizers are
-deployment-specific). Continuous batching and the single-query decode
-kernel are round-2 items (TODO.md).
+Scope: single process (tp=pp=1), token-id interface (tokenizers are
+deployment-specific). The single-query decode kernel is the default
+GPU decode path (round 2); ``--continuous`` serves every request
+through one shared ContinuousBatcher decode loop (requests join and
+leave the running batch between steps) instead of per-request
+generate() calls.
 """
 
 from __future__ import annotations
@@ -25,7 +28,53 @@ from metis_amd.models.gpt import GPTModel, MODEL_SPECS as _GPT_SPECS
 from metis_amd.models.llama import LlamaModel, LlamaModelSpec, LLAMA_SPECS
 
 MODEL_SPECS = {**_GPT_SPECS, **LLAMA_SPECS}
-from metis_amd.runtime.generate import generate, generate_ragged  # noqa: E402
+from metis_amd.runtime.generate import (ContinuousBatcher, generate,  # noqa: E402
+                                         generate_ragged)
+
+
+class BatcherWorker:
+    """One background decode loop shared by all requests: /generate
+    submits into the ContinuousBatcher and blocks until its request
+    retires; the loop steps whenever work is queued."""
+
+    def __init__(self, model, capacity: int, max_batch: int = 8,
+                 device=None, eos_id=None):
+        import threading
+
+        self.cb = ContinuousBatcher(model, capacity, max_batch=max_batch,
+                                    device=device, eos_id=eos_id)
+        self.lock = threading.Lock()
+        self.events = {}
+        self.results = {}
+        self._threading = threading
+        t = threading.Thread(target=self._loop, daemon=True)
+        t.start()
+
+    def submit_and_wait(self, prompt, max_new, temperature, top_k,
+                        generator=None, timeout=600.0):
+        with self.lock:
+            rid = self.cb.submit(prompt, max_new, temperature=temperature,
+                                 top_k=top_k, generator=generator)
+            ev = self.events[rid] = self._threading.Event()
+        if not ev.wait(timeout):
+            raise TimeoutError("generation timed out")
+        with self.lock:
+            self.events.pop(rid, None)
+            return self.results.pop(rid)
+
+    def _loop(self):
+        import time as _time
+
+        while True:
+            with self.lock:
+                busy = self.cb.active > 0 or bool(self.cb._pending)
+                done = self.cb.step() if busy else {}
+                for rid, toks in done.items():
+                    self.results[rid] = toks
+                    if rid in self.events:
+                        self.events[rid].set()
+            if not busy:
+                _time.sleep(0.002)
 
 
 def load_model(model_name: str, checkpoint: str = None,
@@ -58,10 +107,14 @@ except ImportError:  # pragma: no cover - fastapi/pydantic are optional
     GenRequest = None
 
 
-def build_app(model, spec, device=None):
+def build_app(model, spec, device=None, continuous: bool = False,
+              max_batch: int = 8):
     from fastapi import FastAPI
 
     app = FastAPI(title="metis_amd serve")
+    worker = (BatcherWorker(model, spec.seq_length, max_batch=max_batch,
+                            device=device) if continuous else None)
+    app.state.worker = worker
 
     @app.get("/health")
     def health():
@@ -81,6 +134,17 @@ def build_app(model, spec, device=None):
                 400, f"prompt length {max(lens)} >= context {spec.seq_length}")
         budget = spec.seq_length - max(lens)
         n = max(0, min(req.max_new_tokens, budget))
+        if worker is not None:
+            # continuous batching: each prompt becomes one request in
+            # the shared decode loop (requests from other HTTP calls
+            # interleave in the same batch)
+            g = None
+            if req.seed:
+                g = torch.Generator().manual_seed(req.seed)
+            outs = [worker.submit_and_wait(p, n, req.temperature,
+                                           req.top_k, generator=g)
+                    for p in req.tokens]
+            return {"tokens": outs}
         if len(lens) > 1:   # ragged batch: padded-cache batched decode
             out = generate_ragged(model, req.tokens, n,
                                   temperature=req.temperature,
@@ -103,6 +167,10 @@ def main() -> None:
     p.add_argument("--checkpoint", default=None)
     p.add_argument("--host", default="127.0.0.1")
     p.add_argument("--port", type=int, default=8000)
+    p.add_argument("--continuous", action="store_true",
+                   help="serve through one shared continuous-batching "
+                        "decode loop")
+    p.add_argument("--max-batch", type=int, default=8)
     args = p.parse_args()
 
     device = "cuda:0" if torch.cuda.is_available() else None
@@ -111,7 +179,8 @@ def main() -> None:
 
     import uvicorn
 
-    uvicorn.run(build_app(model, spec, device), host=args.host,
+    uvicorn.run(build_app(model, spec, device, continuous=args.continuous,
+                          max_batch=args.max_batch), host=args.host,
                 port=args.port, log_level="warning")
 
 

@@ -175,3 +175,137 @@ def generate(
         out = torch.cat([out, nxt], dim=1)
         logits = model(nxt, cache=cache, pos_offset=cache.seq_len)
     return out
+
+
+class ContinuousBatcher:
+    """Continuous batching over the padded ragged cache: requests join
+    and leave the decode batch between steps instead of waiting for a
+    whole batch to finish (the serving pattern vLLM popularized; the
+    reference has no inference path at all).
+
+    Fixed ``max_batch`` slots share per-layer K/V buffers of
+    ``capacity`` positions. ``submit()`` queues a request; each
+    ``step()`` admits pending requests into free slots (one exact
+    batch-1 prefill each, spliced into the slot's buffer rows), decodes
+    ONE token for every active slot, and retires rows that hit their
+    token budget or ``eos_id``. Retired slots are reused immediately —
+    a long request no longer blocks short ones behind it.
+
+    Single process (tp=pp=1), device-agnostic (CPU tests; the GPU path
+    uses the default-on single-query decode kernel via the models).
+    """
+
+    def __init__(self, model, capacity: int, max_batch: int = 8,
+                 device=None, eos_id: Optional[int] = None):
+        self.model = model
+        self.capacity = capacity
+        self.max_batch = max_batch
+        self.device = device
+        self.eos_id = eos_id
+        self.cache: Optional[RaggedKVCache] = None
+        self._pending = []      # submitted, not yet admitted
+        self._slots = [None] * max_batch   # slot -> request state or None
+        self._next_id = 0
+        self._last_logits = {}  # slot -> [vocab] logits of last position
+        self.finished = {}      # req_id -> token list
+
+    def submit(self, prompt, max_new_tokens: int, temperature: float = 0.0,
+               top_k: int = 0, generator=None) -> int:
+        assert len(prompt) > 0
+        assert len(prompt) + max_new_tokens <= self.capacity, (
+            "prompt + budget exceeds cache capacity")
+        rid = self._next_id
+        self._next_id += 1
+        self._pending.append(dict(
+            rid=rid, prompt=list(prompt), max_new=max_new_tokens,
+            temperature=temperature, top_k=top_k, generator=generator,
+            out=list(prompt)))
+        return rid
+
+    @property
+    def active(self) -> int:
+        return sum(s is not None for s in self._slots)
+
+    def _ensure_cache(self, proto_cache: "KVCache") -> None:
+        if self.cache is not None:
+            return
+        self.cache = RaggedKVCache(torch.zeros(self.max_batch,
+                                               dtype=torch.long))
+        self.cache.active = torch.zeros(self.max_batch, dtype=torch.bool)
+        for idx, (k, v) in proto_cache._kv.items():
+            kbuf = k.new_zeros(self.max_batch, k.size(1), self.capacity,
+                               k.size(3))
+            self.cache._kv[idx] = (kbuf, torch.zeros_like(kbuf))
+        if self.device is not None:
+            self.cache.lengths = self.cache.lengths.to(self.device)
+            self.cache.active = self.cache.active.to(self.device)
+
+    def _admit(self) -> None:
+        while self._pending and self.active < self.max_batch:
+            req = self._pending.pop(0)
+            slot = self._slots.index(None)
+            toks = torch.tensor([req["prompt"]], dtype=torch.long,
+                                device=self.device)
+            pre = KVCache()
+            logits = self.model(toks, cache=pre, pos_offset=0)
+            self._ensure_cache(pre)
+            plen = len(req["prompt"])
+            for idx, (k, v) in pre._kv.items():
+                kbuf, vbuf = self.cache._kv[idx]
+                kbuf[slot, :, :plen] = k[0]
+                vbuf[slot, :, :plen] = v[0]
+            self.cache.lengths[slot] = plen
+            self.cache.active[slot] = True
+            self._slots[slot] = req
+            self._last_logits[slot] = logits[0, -1]
+
+    def step(self):
+        """Admit + decode one token for every active row; returns the
+        dict of requests that finished THIS step ({rid: tokens})."""
+        self._admit()
+        done_now = {}
+        if self.active == 0:
+            return done_now
+        # sample per row (per-request params), inactive rows decode a
+        # dummy token that the mask and retirement logic ignore
+        nxt = torch.zeros(self.max_batch, 1, dtype=torch.long,
+                          device=self.device)
+        for slot, req in enumerate(self._slots):
+            if req is None:
+                continue
+            last = self._last_logits[slot][None].float()
+            tok = _sample(last, req["temperature"], req["top_k"],
+                          req["generator"])
+            req["out"].append(int(tok[0, 0]))
+            nxt[slot, 0] = tok[0, 0]
+        # batched single-token forward against the padded cache
+        pos = self.cache.lengths.clone()
+        # inactive rows: write their garbage token at a parked position
+        # (their stale length) — overwritten at the next admit, hidden
+        # by attention_mask() until then; clamp against the buffer end
+        pos = torch.clamp(pos, max=self.capacity - 1)
+        self.cache.lengths = pos
+        logits = self.model(nxt, cache=self.cache, pos_offset=pos)[:, -1]
+        # advance only active rows
+        self.cache.lengths = pos + self.cache.active.long()
+        for slot, req in enumerate(self._slots):
+            if req is None:
+                continue
+            self._last_logits[slot] = logits[slot]
+            tok = req["out"][-1]
+            n_new = len(req["out"]) - len(req["prompt"])
+            if n_new >= req["max_new"] or (self.eos_id is not None
+                                           and tok == self.eos_id):
+                done_now[req["rid"]] = req["out"]
+                self.finished[req["rid"]] = req["out"]
+                self._slots[slot] = None
+                self.cache.active[slot] = False
+                del self._last_logits[slot]
+        return done_now
+
+    def run_until_done(self, max_steps: int = 100000):
+        for _ in range(max_steps):
+            self.step()
+            if self.active == 0 and not self._pending:
+                return self.finished
+        raise RuntimeError("continuous batcher did not drain")

@@ -90,3 +90,57 @@ def test_generate_ragged_matches_per_sequence():
             ref.append(generate(model, toks, 6, temperature=0.0)[0].tolist())
         rag = generate_ragged(model, prompts, 6, temperature=0.0)
         assert rag == ref
+
+
+def test_continuous_batcher_matches_sequential_greedy():
+    """Requests admitted at different times through the continuous
+    batcher produce exactly the sequential greedy continuations."""
+    from metis_amd.runtime.generate import ContinuousBatcher
+
+    torch.manual_seed(1)
+    model = GPTModel(GPT_SPEC, dtype=torch.float32)
+    model.eval()
+    g = torch.Generator().manual_seed(5)
+    prompts = [torch.randint(0, GPT_SPEC.vocab_size, (1, n), generator=g)
+               for n in (5, 9, 3, 7)]
+    budgets = [6, 3, 8, 4]
+
+    expected = {}
+    for i, (p, n) in enumerate(zip(prompts, budgets)):
+        out = generate(model, p, n, temperature=0.0)
+        expected[i] = out[0].tolist()
+
+    cb = ContinuousBatcher(model, capacity=GPT_SPEC.seq_length, max_batch=2)
+    # two requests up front (fills both slots), two submitted later —
+    # they must join as slots free up
+    r0 = cb.submit(prompts[0][0].tolist(), budgets[0])
+    r1 = cb.submit(prompts[1][0].tolist(), budgets[1])
+    cb.step()
+    cb.step()
+    r2 = cb.submit(prompts[2][0].tolist(), budgets[2])
+    cb.step()
+    r3 = cb.submit(prompts[3][0].tolist(), budgets[3])
+    finished = cb.run_until_done()
+
+    assert set(finished) == {r0, r1, r2, r3}
+    for rid, i in zip((r0, r1, r2, r3), range(4)):
+        assert finished[rid] == expected[i], (rid, i)
+
+
+def test_continuous_batcher_eos_and_slot_reuse():
+    from metis_amd.runtime.generate import ContinuousBatcher
+
+    torch.manual_seed(2)
+    model = GPTModel(GPT_SPEC, dtype=torch.float32)
+    model.eval()
+    g = torch.Generator().manual_seed(9)
+    p = torch.randint(0, GPT_SPEC.vocab_size, (1, 4), generator=g)
+    # find what greedy emits first, use it as eos -> retires after 1 token
+    first = int(generate(model, p, 1, temperature=0.0)[0, -1])
+    cb = ContinuousBatcher(model, capacity=GPT_SPEC.seq_length,
+                           max_batch=1, eos_id=first)
+    r0 = cb.submit(p[0].tolist(), 10)
+    r1 = cb.submit(p[0].tolist(), 1)   # waits for the slot
+    out = cb.run_until_done()
+    assert len(out[r0]) == 4 + 1       # stopped at eos, not budget
+    assert len(out[r1]) == 4 + 1

@@ -89,3 +89,40 @@ def test_bad_requests_rejected():
     too_long = list(range(SPEC.seq_length + 1))
     assert client.post(
         "/generate", json={"tokens": [too_long]}).status_code == 400
+
+
+def test_continuous_serving_interleaves_requests():
+    """--continuous mode: concurrent HTTP requests share one decode
+    loop and return the same greedy continuations as direct calls."""
+    import threading
+
+    from metis_amd.cli.serve import build_app
+    from metis_amd.runtime.generate import generate
+    from fastapi.testclient import TestClient
+
+    torch.manual_seed(0)
+    model = GPTModel(SPEC, dtype=torch.float32)
+    model.eval()
+    app = build_app(model, SPEC, continuous=True, max_batch=2)
+    client = TestClient(app)
+
+    g = torch.Generator().manual_seed(21)
+    prompts = [torch.randint(0, SPEC.vocab_size, (1, n), generator=g)
+               for n in (4, 6, 5)]
+    expected = [generate(model, p, 5, temperature=0.0)[0].tolist()
+                for p in prompts]
+
+    results = [None] * 3
+
+    def post(i):
+        r = client.post("/generate", json={
+            "tokens": [prompts[i][0].tolist()], "max_new_tokens": 5})
+        assert r.status_code == 200
+        results[i] = r.json()["tokens"][0]
+
+    threads = [threading.Thread(target=post, args=(i,)) for i in range(3)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=120)
+    assert results == expected
