@@ -144,3 +144,25 @@ def test_continuous_batcher_eos_and_slot_reuse():
     out = cb.run_until_done()
     assert len(out[r0]) == 4 + 1       # stopped at eos, not budget
     assert len(out[r1]) == 4 + 1
+
+
+def test_continuous_batcher_capacity_and_sampling():
+    from metis_amd.runtime.generate import ContinuousBatcher
+    import pytest as _pytest
+
+    torch.manual_seed(3)
+    model = GPTModel(GPT_SPEC, dtype=torch.float32)
+    model.eval()
+    cb = ContinuousBatcher(model, capacity=16, max_batch=2)
+    # over-capacity request rejected up front
+    with _pytest.raises(AssertionError):
+        cb.submit(list(range(10)), 7)
+    # sampled path is deterministic under a fixed generator
+    g1 = torch.Generator().manual_seed(77)
+    g2 = torch.Generator().manual_seed(77)
+    r1 = cb.submit([1, 2, 3], 6, temperature=0.8, top_k=8, generator=g1)
+    out1 = cb.run_until_done()[r1]
+    cb2 = ContinuousBatcher(model, capacity=16, max_batch=2)
+    r2 = cb2.submit([1, 2, 3], 6, temperature=0.8, top_k=8, generator=g2)
+    out2 = cb2.run_until_done()[r2]
+    assert out1 == out2

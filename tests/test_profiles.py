@@ -64,3 +64,33 @@ def test_write_read_roundtrip(tmp_path):
     assert prof.fb_sync_ms == pytest.approx(90.0 - 85.0)
     assert store.model.optimizer_time_ms == 10.0
     assert store.model.parameters_per_layer_bytes == [10.0, 20.0, 10.0]
+
+
+def test_emulated_tp_profile_ingest(tmp_path):
+    """Profiles with the tp_comm_modeled extension keys load like any
+    other (the loader keeps unknown execution_time keys harmless) and
+    still expose the accumulation marginal."""
+    from metis_amd.profiles import ProfileStore
+
+    path = tmp_path / "DeviceType.MI355X_tp2_bs1.json"
+    ProfileStore.write_profile_json(
+        str(path), model_name="m",
+        parameters_per_layer_bytes=[10.0, 20.0, 10.0],
+        total_time_ms=12.0, forward_backward_time_ms=10.0,
+        batch_generator_time_ms=0.1,
+        layernorm_grads_all_reduce_time_ms=0.3,
+        embedding_grads_all_reduce_time_ms=0.2,
+        optimizer_time_ms=1.0,
+        layer_compute_total_ms=[2.0, 5.0, 2.0],
+        total_memory_mb=30.0, layer_memory_total_mb=[10.0, 10.0, 10.0],
+        fwd_bwd_2mb_ms=18.0, fwd_bwd_4mb_ms=34.0,
+        extra_execution_keys={"tp_comm_modeled": True,
+                              "tp_comm_bw_GBps": 130.0,
+                              "tp_comm_alpha_us": 20.0})
+    store = ProfileStore.load_dir(str(tmp_path))
+    prof = store.get("MI355X", 2, 1)
+    assert prof.marginal_mb_ms == 8.0      # (34-18)/2
+    assert prof.residual_ms == 2.0         # 18 - 2*8
+    import json
+    raw = json.load(open(path))
+    assert raw["execution_time"]["tp_comm_modeled"] is True
