@@ -35,6 +35,19 @@ def build_parser() -> argparse.ArgumentParser:
     # MI355X extensions
     p.add_argument("--comm_model", choices=["parity", "alpha_beta"], default="parity")
     p.add_argument("--alpha_us", type=float, default=20.0)
+    p.add_argument("--microbatch_model", choices=["parity", "marginal"],
+                   default="parity",
+                   help="marginal: price accumulated microbatches at the "
+                        "measured fwd_bwd_{2,4}mb probe marginal, iteration "
+                        "residual charged once")
+    p.add_argument("--interpolate_bs", action="store_true",
+                   help="linearly interpolate profile quantities between "
+                        "profiled batch sizes instead of skipping plans")
+    p.add_argument("--schedule", choices=["gpipe", "1f1b", "interleaved"],
+                   default="gpipe",
+                   help="pipeline schedule the homo estimator prices")
+    p.add_argument("--vpp", type=int, default=1,
+                   help="virtual chunks per stage for --schedule interleaved")
     p.add_argument("--activation_dtype_bytes", type=int, default=1)
     p.add_argument("--drop_incomplete_partitions", action="store_true",
                    help="drop plans whose layer partition misses layers "
@@ -67,5 +80,9 @@ def parse(argv: Optional[List[str]] = None):
         alpha_us=args.alpha_us,
         activation_dtype_bytes=args.activation_dtype_bytes,
         drop_incomplete_partitions=args.drop_incomplete_partitions,
+        microbatch_model=args.microbatch_model,
+        interpolate_bs=args.interpolate_bs,
+        schedule=args.schedule,
+        vpp=args.vpp,
     )
     return args, model_config, planner_args
