@@ -45,15 +45,45 @@ def latest_complete_step(ckpt_dir: str) -> int:
     return best
 
 
-def save_step(runner: PlanRunner, ckpt_dir: str, step: int) -> None:
-    d = os.path.join(ckpt_dir, f"step_{step}")
+def save_step(runner: PlanRunner, ckpt_dir: str, step: int,
+              tag: str = None) -> None:
+    """Periodic checkpoint: per-rank shard + a COMPLETE marker written
+    only after a barrier. With ``tag`` (emergency save from the failure
+    handler) the barrier and marker are skipped — peers may be dead,
+    and resume must not trust a possibly-inconsistent snapshot."""
+    name = f"step_{step}" if tag is None else f"step_{step}_{tag}"
+    d = os.path.join(ckpt_dir, name)
     os.makedirs(d, exist_ok=True)
     runner.save_checkpoint(os.path.join(d, f"rank{runner.ctx.rank}.pt"))
+    if tag is not None:
+        return
     if dist.is_initialized():
         dist.barrier()  # marker only after EVERY rank's shard is on disk
     if runner.ctx.rank == 0:
         with open(os.path.join(d, "COMPLETE"), "w") as fh:
             fh.write(json.dumps({"step": step, "time": time.time()}))
+
+
+def _train_one(args, ctx, runner, sched, step, start, t0,
+               tokens_per_step, eval_loader, eval_loss) -> int:
+    import time
+
+    if sched is not None:
+        sched.step(step)
+    loss = runner.train_step()
+    done = step + 1
+    if ctx.rank == ctx.world_size - 1 and done % args.log_every == 0:
+        tps = tokens_per_step * (done - start) / max(time.time() - t0, 1e-9)
+        print(f"step {done}: loss {loss:.4f} lr {runner.optimizer.lr:.2e} "
+              f"{tps:,.0f} tok/s", flush=True)
+    if eval_loader is not None and done % args.eval_every == 0:
+        el = eval_loss()
+        if ctx.rank == ctx.world_size - 1:
+            print(f"step {done}: eval loss {el:.4f}", flush=True)
+    if (args.checkpoint_dir and args.checkpoint_every > 0
+            and done % args.checkpoint_every == 0):
+        save_step(runner, args.checkpoint_dir, done)
+    return done
 
 
 def main() -> None:
@@ -158,23 +188,31 @@ def main() -> None:
 
     t0 = time.time()
     tokens_per_step = args.gbs * MODEL_SPECS[args.model].seq_length
-    for step in range(start, args.steps):
-        if sched is not None:
-            sched.step(step)
-        loss = runner.train_step()
-        done = step + 1
-        if ctx.rank == ctx.world_size - 1 and done % args.log_every == 0:
-            tps = tokens_per_step * (done - start) / max(time.time() - t0, 1e-9)
-            print(f"step {done}: loss {loss:.4f} lr {runner.optimizer.lr:.2e} "
-                  f"{tps:,.0f} tok/s", flush=True)
-        if eval_loader is not None and done % args.eval_every == 0:
-            el = eval_loss()
-            if ctx.rank == ctx.world_size - 1:
-                print(f"step {done}: eval loss {el:.4f}", flush=True)
-        if (args.checkpoint_dir and args.checkpoint_every > 0
-                and done % args.checkpoint_every == 0):
-            save_step(runner, args.checkpoint_dir, done)
+    done = start
+    try:
+        for step in range(start, args.steps):
+            done = _train_one(args, ctx, runner, sched, step, start, t0,
+                              tokens_per_step, eval_loader, eval_loss)
+    except Exception as e:
+        # failure detection: a collective timeout / kernel fault lands
+        # here — record a structured failure marker and (best effort,
+        # the local state may still be intact) an emergency checkpoint
+        # so --checkpoint-dir runs resume from the last good step
+        import json as _json, traceback
 
+        if args.checkpoint_dir:
+            os.makedirs(args.checkpoint_dir, exist_ok=True)
+            with open(os.path.join(args.checkpoint_dir,
+                                   f"FAILED_rank{ctx.rank}.json"), "w") as fh:
+                _json.dump({"rank": ctx.rank, "step": done,
+                            "error": repr(e),
+                            "traceback": traceback.format_exc()}, fh)
+            try:
+                save_step(runner, args.checkpoint_dir, done,
+                          tag="emergency")
+            except Exception:
+                pass
+        raise
     if args.checkpoint_dir and args.steps > start:
         save_step(runner, args.checkpoint_dir, args.steps)
     if dist.is_initialized():

@@ -82,9 +82,13 @@ def init_parallel(
     if world > 1 and not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29531")
+        # failure detection: a hung collective (dead peer, deadlock)
+        # surfaces as a timeout error instead of an infinite hang;
+        # METIS_DIST_TIMEOUT_S tunes it (300 s default)
+        timeout_s = int(os.environ.get("METIS_DIST_TIMEOUT_S", "300"))
         dist.init_process_group(
             backend=backend, rank=rank, world_size=world,
-            timeout=datetime.timedelta(seconds=300),
+            timeout=datetime.timedelta(seconds=timeout_s),
         )
 
     ctx = ParallelContext(rank=rank, world_size=world, local_rank=local_rank,

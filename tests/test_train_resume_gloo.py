@@ -108,3 +108,39 @@ train.main()
     assert r.returncode == 0, r.stderr[-1500:]
     assert "step 1:" in r.stdout          # log-every from yaml
     assert "step 2" not in r.stdout       # --steps 1 overrode yaml's 2
+
+
+def test_failure_marker_and_emergency_checkpoint(tmp_path, monkeypatch):
+    """A mid-run exception writes FAILED_rank json + an emergency
+    checkpoint (no COMPLETE marker -> resume ignores it) and re-raises."""
+    import sys
+
+    from metis_amd.cli import train as train_mod
+    from metis_amd.cli.train import latest_complete_step
+
+    calls = {"n": 0}
+    orig = train_mod.PlanRunner.train_step
+
+    def boom(self):
+        calls["n"] += 1
+        if calls["n"] >= 3:
+            raise RuntimeError("injected collective failure")
+        return orig(self)
+
+    monkeypatch.setattr(train_mod.PlanRunner, "train_step", boom)
+    monkeypatch.setattr(sys, "argv", [
+        "train", "--model", "gpt2-small", "--steps", "5", "--mbs", "1",
+        "--gbs", "1", "--checkpoint-dir", str(tmp_path),
+        "--checkpoint-every", "2"])
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError, match="injected"):
+        train_mod.main()
+
+    import json as _json
+    marker = tmp_path / "FAILED_rank0.json"
+    assert marker.exists()
+    doc = _json.loads(marker.read_text())
+    assert doc["step"] == 2 and "injected" in doc["error"]
+    assert (tmp_path / "step_2_emergency" / "rank0.pt").exists()
+    # resume only trusts COMPLETE checkpoints (step_2 has the marker)
+    assert latest_complete_step(str(tmp_path)) == 2
