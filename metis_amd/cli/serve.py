@@ -50,17 +50,25 @@ class BatcherWorker:
         t = threading.Thread(target=self._loop, daemon=True)
         t.start()
 
-    def submit_and_wait(self, prompt, max_new, temperature, top_k,
-                        generator=None, timeout=600.0):
+    def submit(self, prompt, max_new, temperature, top_k, generator=None):
         with self.lock:
             rid = self.cb.submit(prompt, max_new, temperature=temperature,
                                  top_k=top_k, generator=generator)
-            ev = self.events[rid] = self._threading.Event()
+            self.events[rid] = self._threading.Event()
+        return rid
+
+    def wait(self, rid, timeout=600.0):
+        ev = self.events[rid]
         if not ev.wait(timeout):
             raise TimeoutError("generation timed out")
         with self.lock:
             self.events.pop(rid, None)
             return self.results.pop(rid)
+
+    def submit_and_wait(self, prompt, max_new, temperature, top_k,
+                        generator=None, timeout=600.0):
+        return self.wait(self.submit(prompt, max_new, temperature, top_k,
+                                     generator), timeout)
 
     def _loop(self):
         import time as _time
@@ -141,10 +149,11 @@ def build_app(model, spec, device=None, continuous: bool = False,
             g = None
             if req.seed:
                 g = torch.Generator().manual_seed(req.seed)
-            outs = [worker.submit_and_wait(p, n, req.temperature,
-                                           req.top_k, generator=g)
-                    for p in req.tokens]
-            return {"tokens": outs}
+            # submit every prompt first so they decode as ONE batch,
+            # then collect
+            rids = [worker.submit(p, n, req.temperature, req.top_k,
+                                  generator=g) for p in req.tokens]
+            return {"tokens": [worker.wait(r) for r in rids]}
         if len(lens) > 1:   # ragged batch: padded-cache batched decode
             out = generate_ragged(model, req.tokens, n,
                                   temperature=req.temperature,
