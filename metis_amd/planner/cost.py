@@ -261,7 +261,16 @@ class HeteroCostEstimator(CostEstimatorBase):
             dp_costs.append(self._dp_cost([stage_params], dp_bw, dp_deg))
             opt_costs.append(self._optimizer_cost(tp_deg, end_l - start_l))
 
-        execution = (plan.batches - 1) * max(lens) + sum(lens)
+        # schedule pricing mirrors HomoCostEstimator: interleaved (vpp=v)
+        # shrinks the bubble to (pp-1)/v chunk slots at v x the p2p
+        # volume; gpipe/1f1b keep the reference bubble
+        sched, v = self.args.schedule, max(self.args.vpp, 1)
+        if sched == "interleaved" and plan.num_stage > 1:
+            execution = ((plan.batches - 1) * max(lens)
+                         + (sum(lens) + (v - 1) * max(lens)) / v)
+            pp_cost *= v
+        else:
+            execution = (plan.batches - 1) * max(lens) + sum(lens)
         total = (
             execution + fb_sync_cost + max(opt_costs) + max(dp_costs) + pp_cost
             + self._batch_gen_cost(plan.batches)
