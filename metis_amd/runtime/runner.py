@@ -125,6 +125,8 @@ class PlanRunner:
         self.dtype = dtype
         self.tracer = tracer_from_env(ctx.rank)
         self._check_sync = os.environ.get("METIS_CHECK_SYNC") == "1"
+        _sw = os.environ.get("METIS_STRAGGLER_WARN")
+        self._straggler_warn = float(_sw) if _sw else 0.0
         self._data_gen = None
         # real-data path: memory-mapped token file with the same
         # determinism contract as the synthetic stream (data/dataset.py)
@@ -498,6 +500,12 @@ class PlanRunner:
         return global_grad_norm(pairs, self.ctx.tp_group, self.ctx.pp_group)
 
     def _sync_and_step(self) -> None:
+        if getattr(self, "_step_t0", None) is not None:
+            # straggler detector: compute-segment end (collectives that
+            # equalize ranks come after this point)
+            if (self.ctx.device or torch.device("cpu")).type == "cuda":
+                torch.cuda.synchronize()
+            self._compute_s = time.perf_counter() - self._step_t0
         with self.tracer.span("grad_sync"):
             if self.grad_sync is not None:
                 # hooks copied + all-reduced the final-microbatch grads,
@@ -539,13 +547,40 @@ class PlanRunner:
             )
 
     def train_step(self) -> float:
+        self._step_t0 = time.perf_counter() if self._straggler_warn else None
         if self.ctx.pp == 1:
-            return self._step_no_pipeline()
-        if self.schedule == "interleaved":
-            return self._step_interleaved()
-        if self.schedule == "1f1b":
-            return self._step_pipeline_1f1b()
-        return self._step_pipeline()
+            loss = self._step_no_pipeline()
+        elif self.schedule == "interleaved":
+            loss = self._step_interleaved()
+        elif self.schedule == "1f1b":
+            loss = self._step_pipeline_1f1b()
+        else:
+            loss = self._step_pipeline()
+        if self._step_t0 is not None:
+            self._check_stragglers()
+        return loss
+
+    def _check_stragglers(self) -> None:
+        """Straggler detector (METIS_STRAGGLER_WARN=<ratio>): all-gather
+        per-rank COMPUTE segment times (step start to _sync_and_step —
+        whole-step wall times equalize at the blocking collectives, the
+        pre-sync segment is where a throttled or failing GPU shows up);
+        rank 0 warns when max/min exceeds the ratio. Synchronizing, so
+        it is opt-in diagnostics, not an always-on cost."""
+        if not dist.is_initialized() or self.ctx.world_size < 2:
+            return
+        dev = self.ctx.device or torch.device("cpu")
+        t = torch.tensor([self._compute_s], dtype=torch.float64, device=dev)
+        gathered = [torch.zeros_like(t) for _ in range(self.ctx.world_size)]
+        dist.all_gather(gathered, t)
+        times = [float(g) for g in gathered]
+        lo = max(min(times), 1e-9)
+        if self.ctx.rank == 0 and max(times) / lo > self._straggler_warn:
+            worst = times.index(max(times))
+            print(f"WARNING: rank {worst} compute time "
+                  f"{max(times)*1e3:.1f} ms vs fastest {lo*1e3:.1f} ms "
+                  f"({max(times)/lo:.2f}x > {self._straggler_warn:.2f}x "
+                  "threshold)", flush=True)
 
     # --- checkpoint ---------------------------------------------------------
     def save_checkpoint(self, path: str) -> None:

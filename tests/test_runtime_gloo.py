@@ -519,3 +519,37 @@ def _sp_pp_worker(rank, world, port, out):
 
 def test_sp_with_pipeline_boundaries():
     _run_workers(_sp_pp_worker, world=4, port=29627)
+
+
+def _straggler_worker(rank, world, port, out):
+    _env(rank, world, port)
+    import io
+    import time as _time
+    from contextlib import redirect_stdout
+
+    os.environ["METIS_STRAGGLER_WARN"] = "1.5"
+    from metis_amd.runtime.comm import init_parallel
+    from metis_amd.runtime.runner import PlanRunner
+
+    ctx = init_parallel(dp=2, tp=1, pp=1)
+    torch.manual_seed(0)
+    runner = PlanRunner(SPEC, ctx, mbs=1, gbs=2, dtype=torch.float32)
+    orig = runner._step_no_pipeline
+
+    def slow():
+        if ctx.rank == 1:
+            _time.sleep(0.4)
+        return orig()
+
+    runner._step_no_pipeline = slow
+    buf = io.StringIO()
+    with redirect_stdout(buf):
+        runner.train_step()
+    if ctx.rank == 0:
+        assert "WARNING: rank 1" in buf.getvalue(), buf.getvalue()
+    out.put(("ok", rank))
+    dist.destroy_process_group()
+
+
+def test_straggler_detector_warns():
+    _run_workers(_straggler_worker, world=2, port=29640)
