@@ -81,6 +81,8 @@ class BatcherWorker:
                     self.results[rid] = toks
                     if rid in self.events:
                         self.events[rid].set()
+                self.cb.finished.clear()   # served via results; don't
+                                           # accumulate forever
             if not busy:
                 _time.sleep(0.002)
 
