@@ -127,6 +127,16 @@ class PlanRunner:
         self._check_sync = os.environ.get("METIS_CHECK_SYNC") == "1"
         _sw = os.environ.get("METIS_STRAGGLER_WARN")
         self._straggler_warn = float(_sw) if _sw else 0.0
+        # METIS_HIPGRAPH=1: capture the pp==1 fwd+bwd as one hipGraph
+        # (torch.cuda.make_graphed_callables) and replay it per
+        # microbatch — removes the per-kernel launch overhead that
+        # dominates small-model small-mbs steps. Opt-in; requires
+        # static shapes (always true here) and no cross-rank hooks
+        # inside the graph (dp=1 or bucket hooks fire post-backward).
+        self._use_hipgraph = (os.environ.get("METIS_HIPGRAPH") == "1"
+                              and ctx.pp == 1 and ctx.tp == 1
+                              and torch.cuda.is_available())
+        self._graphed = None
         self._data_gen = None
         # real-data path: memory-mapped token file with the same
         # determinism contract as the synthetic stream (data/dataset.py)
@@ -175,6 +185,17 @@ class PlanRunner:
         return self.synthetic_batch()
 
     # --- single-stage step (pp == 1) --------------------------------------
+    def _forward(self, tokens, labels):
+        if self._use_hipgraph:
+            if self._graphed is None:
+                # capture once at first use (shapes are static); the
+                # graphed callable copies fresh inputs into its static
+                # buffers and replays fwd (and bwd on .backward())
+                self._graphed = torch.cuda.make_graphed_callables(
+                    self.model, (tokens, labels))
+            return self._graphed(tokens, labels)
+        return self.model(tokens, labels=labels)
+
     def _step_no_pipeline(self) -> float:
         losses = []
         self.optimizer.zero_grad()
@@ -183,7 +204,7 @@ class PlanRunner:
                 self.grad_sync.arm()
             tokens, labels = self.next_batch()
             with self.tracer.span("forward"):
-                loss = self.model(tokens, labels=labels)
+                loss = self._forward(tokens, labels)
             with self.tracer.span("backward"):
                 (loss / self.num_microbatches).backward()
             losses.append(loss.detach())
