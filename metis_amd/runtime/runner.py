@@ -129,10 +129,11 @@ class PlanRunner:
         self._straggler_warn = float(_sw) if _sw else 0.0
         # METIS_HIPGRAPH=1: capture the pp==1 fwd+bwd as one hipGraph
         # (torch.cuda.make_graphed_callables) and replay it per
-        # microbatch — removes the per-kernel launch overhead that
-        # dominates small-model small-mbs steps. Opt-in; requires
-        # static shapes (always true here) and no cross-rank hooks
-        # inside the graph (dp=1 or bucket hooks fire post-backward).
+        # microbatch. Capture-validated on MI355X; measured SLOWER at
+        # gpt2-small mbs1 (41.5 vs 39.0 ms/step — replay floor + input
+        # copies beat the launch savings), so opt-in and off
+        # (BENCHMARKS.md "hipGraph capture"). Requires static shapes
+        # (always true here); tp=1 keeps collectives out of the graph.
         self._use_hipgraph = (os.environ.get("METIS_HIPGRAPH") == "1"
                               and ctx.pp == 1 and ctx.tp == 1
                               and torch.cuda.is_available())
