@@ -283,3 +283,44 @@ def test_moe_volume_parameter_split():
     assert dense.parameter_sizes(2)[0] == 500.0
     # stage sum consistent
     assert vol.stage_parameter_size(2, 0, 4) == pytest.approx(sum(sizes))
+
+
+def test_hetero_marginal_and_interleaved(marginal_setup):
+    """Hetero estimator under the marginal microbatch model and the
+    interleaved schedule (hand-computed on the 4-GPU 2-node fixture)."""
+    from metis_amd.planner.cost import HeteroCostEstimator
+    from metis_amd.planner.plans import InterStagePlan
+
+    (cluster, _cluster1), store, cfg, vol = marginal_setup
+    plan = InterStagePlan(
+        ns_idx=0, node_sequence=cluster.unique_device_types(),
+        dg_idx=0, device_groups=[2, 2], num_stage=2, batches=2, gbs=4)
+    rank_map = {r: "MI355X" for r in range(4)}
+    strategies = [(2, 1), (2, 1)]          # mbs = 4/2/2 = 1 per stage
+
+    est = HeteroCostEstimator(store, cfg, vol, cluster,
+                              PlannerArgs(gbs=4, max_profiled_tp_degree=1,
+                                          max_profiled_batch_size=4,
+                                          microbatch_model="marginal"))
+    cost = est.get_cost(plan, strategies, [0, 2, 4], rank_map)
+    # marginal(bs1)=7 split [3/8, 5/8] -> lens [2.625, 4.375]
+    lens = [7 * 3 / 8, 7 * 5 / 8]
+    execution = (2 - 1) * max(lens) + sum(lens)
+    residual = 2.0 * 3 / 8 + 2.0 * 5 / 8   # once per stage slice
+    opt = 4.0 * (2 / 4)                    # optimizer * layer ratio / tp
+    pp = 128 / (10 * 1024 * 1024)          # stage boundary spans nodes
+    dp = 2 * 1 / (2 * 100 * 1024 * 1024) * 150
+    bg = 0.5 * 2
+    assert cost == pytest.approx(execution + residual + opt + pp + dp + bg,
+                                 rel=1e-6)
+
+    est_i = HeteroCostEstimator(store, cfg, vol, cluster,
+                                PlannerArgs(gbs=4, max_profiled_tp_degree=1,
+                                            max_profiled_batch_size=4,
+                                            microbatch_model="marginal",
+                                            schedule="interleaved", vpp=2))
+    cost_i = est_i.get_cost(plan, strategies, [0, 2, 4], rank_map)
+    exec_i = (2 - 1) * max(lens) + (sum(lens) + (2 - 1) * max(lens)) / 2
+    assert cost_i == pytest.approx(exec_i + residual + opt + 2 * pp + dp + bg,
+                                   rel=1e-6)
+    assert cost_i < cost                   # smaller bubble wins here
