@@ -40,6 +40,7 @@ torch::Tensor gemm_bf16(
     torch::Tensor a, torch::Tensor b_nk, c10::optional<torch::Tensor> bias,
     long epilogue);
 torch::Tensor mfma_tile_probe(torch::Tensor a, torch::Tensor b);
+torch::Tensor gemm8_bf16(torch::Tensor x, torch::Tensor w);
 std::vector<torch::Tensor> rmsnorm_fwd(
     torch::Tensor x, torch::Tensor gamma, double eps);
 std::vector<torch::Tensor> rmsnorm_bwd(
@@ -75,6 +76,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gemm_bf16", &gemm_bf16,
           "MFMA bf16 GEMM A[M,K] @ B[N,K]^T with fused epilogue "
           "(0=none, 1=bias, 2=bias+gelu)");
+    m.def("gemm8_bf16", &gemm8_bf16,
+          "256^2-tile 2-buffer glds bf16 GEMM x[M,K] @ w[N,K]^T");
     m.def("mfma_tile_probe", &mfma_tile_probe,
           "single-wave 16x16x32 MFMA layout probe");
     m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused RMSNorm forward");

@@ -24,6 +24,7 @@ sources = [
     os.path.join(HIP_DIR, "attention.hip"),
     os.path.join(HIP_DIR, "attention_bwd.hip"),
     os.path.join(HIP_DIR, "gemm.hip"),
+    os.path.join(HIP_DIR, "gemm8.hip"),
     os.path.join(HIP_DIR, "rmsnorm.hip"),
     os.path.join(HIP_DIR, "rope.hip"),
     os.path.join(HIP_DIR, "swiglu.hip"),
